@@ -1,0 +1,431 @@
+"""Distributed K-FAC preconditioner base class.
+
+Keeps the reference's public contract
+(reference: kfac/kfac_preconditioner_base.py:13-231):
+
+* ``optim.Optimizer`` subclass so ``LambdaLR`` / ``KFACParamScheduler``
+  drive ``lr``/``damping``/``*_update_freq`` through ``param_groups``.
+* forward-pre / full-backward hooks on every ``Linear``/``Conv2d`` save
+  activations and output-gradients by reference (zero-copy until
+  ``step()`` consumes them).
+* the 4-phase ``step()`` template: compute/communicate factors ->
+  compute/communicate inverses -> compute/communicate preconditioned
+  gradients -> in-place gradient update with kl-clip; ``exclude_parts``
+  ablation flags skip individual phases for time breakdowns.
+
+MI355X-first changes (not in the reference):
+
+* All per-layer state lives in :class:`FlatBucket` views so each
+  communication phase is one (or per-owner one) large collective instead
+  of a burst of per-layer <=1 MB messages -- the xGMI links are
+  point-to-point and latency-bound for small tensors
+  (SURVEY.md S5 'Distributed communication backend').
+* Owner-rooted broadcasts are issued concurrently on rotating duplicate
+  process groups (separate RCCL comms/streams per group).
+* kl-clip is computed without any host-device sync (the reference calls
+  ``.item()`` per layer, kfac/kfac_preconditioner_inv.py:198-200).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.optim as optim
+
+import kfac_pytorch_amd.parallel.comm as comm_mod
+from kfac_pytorch_amd.parallel.comm import FlatBucket
+
+logger = logging.getLogger(__name__)
+
+SUPPORTED_MODULES = ("Linear", "Conv2d")
+
+
+class KFACBase(optim.Optimizer):
+    """Base distributed K-FAC gradient preconditioner.
+
+    Args mirror the reference (kfac/kfac_preconditioner_base.py:53-77):
+      model, lr, damping, fac_update_freq, kfac_update_freq,
+      communicate_inverse_or_not, kl_clip, factor_decay,
+      exclude_vocabulary_size, hook_enabled, exclude_parts.
+    """
+
+    def __init__(self,
+                 model: nn.Module,
+                 lr: float = 0.1,
+                 damping: float = 0.001,
+                 fac_update_freq: int = 1,
+                 kfac_update_freq: int = 1,
+                 communicate_inverse_or_not: bool = True,
+                 kl_clip: float = 0.001,
+                 factor_decay: float = 0.95,
+                 exclude_vocabulary_size: Optional[int] = None,
+                 hook_enabled: bool = True,
+                 exclude_parts: str = ''):
+
+        defaults = dict(lr=lr, damping=damping,
+                        fac_update_freq=fac_update_freq,
+                        kfac_update_freq=kfac_update_freq)
+        super().__init__(model.parameters(), defaults)
+
+        self.lr = lr
+        self.damping = damping
+        self.fac_update_freq = fac_update_freq
+        self.kfac_update_freq = kfac_update_freq
+        self.communicate_inverse_or_not = communicate_inverse_or_not
+        self.kl_clip = kl_clip if (kl_clip is not None and kl_clip > 0) else None
+        self.factor_decay = factor_decay
+        self.exclude_vocabulary_size = exclude_vocabulary_size
+        self.hook_enabled = hook_enabled
+
+        self.exclude_communicate_inverse = 'CommunicateInverse' in exclude_parts
+        self.exclude_compute_inverse = 'ComputeInverse' in exclude_parts
+        self.exclude_communicate_factor = 'CommunicateFactor' in exclude_parts
+        self.exclude_compute_factor = 'ComputeFactor' in exclude_parts
+
+        self.modules: List[nn.Module] = []
+        self.module_names: List[str] = []
+        self._hook_handles = []
+        self._register_module_hooks(model)
+
+        # per-module saved activations / output-grads (cleared every step)
+        self.m_a: Dict[nn.Module, torch.Tensor] = {}
+        self.m_g: Dict[nn.Module, torch.Tensor] = {}
+
+        # per-module state views (into FlatBuckets, filled by _init_state)
+        self.m_A: Dict[nn.Module, torch.Tensor] = {}
+        self.m_G: Dict[nn.Module, torch.Tensor] = {}
+        self.m_precon_grad: Dict[nn.Module, torch.Tensor] = {}
+
+        self.module_ranks: Optional[Dict[nn.Module, Tuple[int, int]]] = None
+
+        self.eps = 1e-10  # eigenvalue clamp (reference :115)
+        self.steps = 0
+        self._state_ready = False
+
+    # ------------------------------------------------------------------ comm
+    @property
+    def comm(self):
+        return comm_mod.get_comm()
+
+    # ----------------------------------------------------------------- hooks
+    def set_hook_enabled(self, mode: bool = True):
+        self.hook_enabled = mode
+
+    def _save_input_enabled(self) -> bool:
+        return (self.hook_enabled and torch.is_grad_enabled()
+                and self.steps % self.fac_update_freq == 0)
+
+    def _save_grad_enabled(self) -> bool:
+        return self.hook_enabled and self.steps % self.fac_update_freq == 0
+
+    def _forward_hook_event(self, module, input):
+        """Save the module input (a) by reference."""
+        if self._save_input_enabled():
+            self.m_a[module] = input[0].data
+
+    def _backward_hook_event(self, module, grad_input, grad_output):
+        """Save the grad wrt output (g) by reference."""
+        if self._save_grad_enabled():
+            self.m_g[module] = grad_output[0].data
+
+    def _register_module_hooks(self, model: nn.Module):
+        name_idx = 0
+        for module in model.modules():
+            classname = module.__class__.__name__
+            if classname not in SUPPORTED_MODULES:
+                continue
+            if (self.exclude_vocabulary_size is not None
+                    and classname == 'Linear'
+                    and module.out_features == self.exclude_vocabulary_size):
+                continue  # exclude pre-softmax vocab projection (ref :139-140)
+            if classname == 'Conv2d' and module.groups != 1:
+                continue  # grouped conv factors don't match weight layout
+            self.modules.append(module)
+            self._hook_handles.append(
+                module.register_forward_pre_hook(self._forward_hook_event))
+            self._hook_handles.append(
+                module.register_full_backward_hook(self._backward_hook_event))
+            self.module_names.append(f'module_name_{classname}_{name_idx}')
+            name_idx += 1
+        if comm_mod.is_initialized() and self.comm.rank() == 0:
+            logger.info("#register modules: %s", len(self.modules))
+
+    # ------------------------------------------------------------- scheduling
+    def schedule_module_ranks(self):
+        raise NotImplementedError
+
+    def _round_robin_ranks(self, factor_wise: bool = False):
+        """Round-robin layer->rank assignment
+        (reference: kfac/kfac_preconditioner_inv.py:62-77,
+        kfac_preconditioner_eigen.py:75-94 for the factor-wise variant)."""
+        module_ranks = {}
+        size = self.comm.size()
+        rank_iter = 0
+        for module in self.modules:
+            rank_a = rank_iter % size
+            if factor_wise:
+                rank_iter += 1
+                rank_g = rank_iter % size
+            else:
+                rank_g = rank_a
+            module_ranks[module] = (rank_a, rank_g)
+            rank_iter += 1
+        self.module_ranks = module_ranks
+        if self.comm.rank() == 0:
+            logger.info('module_ranks: %s', list(module_ranks.values()))
+        return module_ranks
+
+    # ------------------------------------------------------- state allocation
+    def _init_state(self):
+        """Allocate flat-bucketed per-module state. Subclasses extend."""
+        raise NotImplementedError
+
+    def _reference_param(self) -> torch.Tensor:
+        return self.param_groups[0]['params'][0]
+
+    def _state_device(self) -> torch.device:
+        return self._reference_param().device
+
+    # ----------------------------------------------- flat-bucket allocation
+    def _alloc_factor_buckets(self, owner_only: bool = False):
+        """Allocate m_A/m_G as views of ONE flat fp32 buffer (identity-
+        initialized, reference inits A=G=I at step 0:
+        kfac/kfac_preconditioner_inv.py:84-90).
+
+        ``owner_only=True`` (DP variants) allocates only the factors this
+        rank owns -- DP-KFAC never communicates factors so nobody else
+        needs storage (kfac/kfac_preconditioner_inv_dp.py:75-95).
+        """
+        from kfac_pytorch_amd.ops.factors import factor_dims
+        device = self._state_device()
+        rank = self.comm.rank()
+        self.factor_bucket = FlatBucket(torch.float32, device)
+        for i, m in enumerate(self.modules):
+            rank_a, rank_g = self.module_ranks[m]
+            da, dg = factor_dims(m)
+            if not owner_only or rank == rank_a:
+                self.factor_bucket.add(f"A{i}", torch.Size((da, da)))
+            if not owner_only or rank == rank_g:
+                self.factor_bucket.add(f"G{i}", torch.Size((dg, dg)))
+        self.factor_bucket.freeze(0.0)
+        for i, m in enumerate(self.modules):
+            if f"A{i}" in self.factor_bucket:
+                v = self.factor_bucket.view(f"A{i}")
+                v.diagonal().fill_(1.0)
+                self.m_A[m] = v
+            if f"G{i}" in self.factor_bucket:
+                v = self.factor_bucket.view(f"G{i}")
+                v.diagonal().fill_(1.0)
+                self.m_G[m] = v
+
+    def _alloc_owner_buckets(self, specs) -> List[FlatBucket]:
+        """Build one FlatBucket per owner rank from
+        ``specs = [(name, shape, owner_rank), ...]`` and return the list.
+        Every rank allocates every bucket (broadcast destinations); views
+        are fetched via ``_owner_view``."""
+        device = self._state_device()
+        size = self.comm.size()
+        buckets = [FlatBucket(torch.float32, device) for _ in range(size)]
+        for name, shape, owner in specs:
+            buckets[owner].add(name, torch.Size(shape))
+        for b in buckets:
+            b.freeze(0.0)
+        return buckets
+
+    @staticmethod
+    def _owner_view(buckets: List[FlatBucket], name: str) -> torch.Tensor:
+        for b in buckets:
+            if name in b:
+                return b.view(name)
+        raise KeyError(name)
+
+    # -------------------------------------------------------- bucketed comm
+    def _allreduce_bucket_avg(self, bucket: FlatBucket):
+        """One allreduce-average for a whole phase's tensors."""
+        if bucket.buffer is None or bucket.buffer.numel() == 0:
+            return
+        self.comm.allreduce(bucket.buffer, op=self.comm.Average)
+
+    def _broadcast_owner_buckets(self, buckets: List[FlatBucket]):
+        """Async broadcast of each owner's flat bucket from its rank,
+        issued concurrently on rotating duplicate process groups so
+        different roots ride different xGMI links; then drain.
+
+        Replaces the reference's per-layer broadcast bursts
+        (kfac/kfac_preconditioner_inv.py:132-142,164-175)."""
+        c = self.comm
+        c.ensure_rotating_groups()
+        handles = []
+        for r, b in enumerate(buckets):
+            if b.buffer is None or len(b) == 0:
+                continue
+            handles.append(c.broadcast_async_(b.buffer, src=r,
+                                              group=c.rotating_group(r)))
+        c.synchronize(handles)
+
+    # ------------------------------------------------ phase methods (virtual)
+    def _compute_factors(self):
+        raise NotImplementedError
+
+    def _communicate_factors(self):
+        raise NotImplementedError
+
+    def _compute_inverse(self):
+        raise NotImplementedError
+
+    def _communicate_inverse(self):
+        raise NotImplementedError
+
+    def _compute_pred(self):
+        raise NotImplementedError
+
+    def _communicate_pred(self):
+        raise NotImplementedError
+
+    # --------------------------------------------------------- grad plumbing
+    def _get_grad(self, module: nn.Module) -> torch.Tensor:
+        """Gradient as [out_dim, in_dim(+1)] (reference
+        kfac/kfac_preconditioner_inv.py:145-154)."""
+        if isinstance(module, nn.Conv2d):
+            grad = module.weight.grad.data.view(
+                module.weight.grad.data.size(0), -1)
+        else:
+            grad = module.weight.grad.data
+        if module.bias is not None:
+            grad = torch.cat([grad, module.bias.grad.data.view(-1, 1)], 1)
+        return grad
+
+    def _reshape_preconditioned_grad(self, module, v: torch.Tensor):
+        """Split [out, in(+1)] back into weight/bias shapes (reference
+        kfac/kfac_preconditioner_inv.py:178-186)."""
+        if module.bias is not None:
+            vw = v[:, :-1].reshape(module.weight.grad.data.size())
+            vb = v[:, -1:].reshape(module.bias.grad.data.size())
+            return [vw, vb]
+        return [v.reshape(module.weight.grad.data.size())]
+
+    def _update_grad_in_place(self):
+        """Copy preconditioned grads into .grad and apply the kl-clip
+        rescale nu = min(1, sqrt(kl_clip/|sum(v*g*lr^2)|))
+        (reference: kfac/kfac_preconditioner_inv.py:188-217) with no
+        host-device sync: vg_sum stays a device scalar."""
+        use_clip = self.kl_clip is not None
+        vg_sum = None
+        grads: List[torch.Tensor] = []
+        for module in self.modules:
+            v = self._reshape_preconditioned_grad(
+                module, self.m_precon_grad[module])
+            params = [module.weight] + (
+                [module.bias] if module.bias is not None else [])
+            for p, vi in zip(params, v):
+                if use_clip:
+                    contrib = (vi * p.grad.data).sum() * (self.lr ** 2)
+                    vg_sum = contrib if vg_sum is None else vg_sum + contrib
+                p.grad.data.copy_(vi)
+                grads.append(p.grad.data)
+
+        if use_clip:
+            if self.exclude_communicate_inverse:
+                return  # nu == 1 (reference :209-212)
+            nu = torch.clamp(
+                (self.kl_clip / vg_sum.abs().clamp_min(1e-30)).sqrt(),
+                max=1.0)
+            torch._foreach_mul_(grads, nu)
+
+    # ------------------------------------------------------------------ step
+    def step(self, closure=None, epoch=None):
+        """One K-FAC step (4-phase template, reference
+        kfac/kfac_preconditioner_base.py:185-230)."""
+        group = self.param_groups[0]
+        self.lr = group['lr']
+        self.damping = group['damping']
+        self.fac_update_freq = group['fac_update_freq']
+        self.kfac_update_freq = group['kfac_update_freq']
+
+        if self.module_ranks is None:
+            self.schedule_module_ranks()
+        if not self._state_ready:
+            self._init_state()
+            self._state_ready = True
+
+        if self.steps % self.fac_update_freq == 0:
+            if not self.exclude_compute_factor:
+                self._compute_factors()
+            if not self.exclude_communicate_factor and self.comm.size() > 1:
+                self._communicate_factors()
+
+        if self.steps % self.kfac_update_freq == 0:
+            if not self.exclude_compute_inverse:
+                self._compute_inverse()
+            if (not self.exclude_communicate_inverse
+                    and self.communicate_inverse_or_not
+                    and self.comm.size() > 1):
+                self._communicate_inverse()
+
+        if not self.exclude_compute_inverse:
+            self._compute_pred()
+
+        if (not self.exclude_communicate_inverse
+                and not self.communicate_inverse_or_not
+                and self.comm.size() > 1):
+            self._communicate_pred()
+
+        if not self.exclude_compute_inverse:
+            self._update_grad_in_place()
+
+        self.steps += 1
+        self.m_a, self.m_g = {}, {}
+
+
+class KFACParamScheduler:
+    """Epoch-schedule for damping and update frequencies
+    (reference: kfac/kfac_preconditioner_base.py:233-301)."""
+
+    def __init__(self, kfac, damping_alpha=1, damping_schedule=None,
+                 update_freq_alpha=1, update_freq_schedule=None,
+                 start_epoch=0):
+        self.kfac = kfac
+        params = self.kfac.param_groups[0]
+        self.damping_base = params['damping']
+        self.damping_alpha = damping_alpha
+        self.damping_schedule = damping_schedule
+        self.damping_factor_func = self._get_factor_func(
+            damping_schedule, damping_alpha)
+        self.fac_update_freq_base = params['fac_update_freq']
+        self.kfac_update_freq_base = params['kfac_update_freq']
+        self.update_freq_alpha = update_freq_alpha
+        self.update_freq_schedule = update_freq_schedule
+        self.update_freq_factor_func = self._get_factor_func(
+            update_freq_schedule, update_freq_alpha)
+        self.epoch = start_epoch
+
+    @staticmethod
+    def _get_factor_func(schedule, alpha):
+        if schedule is not None:
+            schedule = sorted(schedule, reverse=True)
+        else:
+            schedule = []
+
+        def factor_func(epoch):
+            factor = 1.0
+            for e in schedule:
+                if epoch >= e:
+                    factor *= alpha
+            return factor
+
+        return factor_func
+
+    def step(self, epoch=None):
+        if epoch is not None:
+            self.epoch = epoch
+        else:
+            self.epoch += 1
+        params = self.kfac.param_groups[0]
+        params['damping'] = self.damping_base * \
+            self.damping_factor_func(self.epoch)
+        factor = self.update_freq_factor_func(self.epoch)
+        params['fac_update_freq'] = int(self.fac_update_freq_base * factor)
+        params['kfac_update_freq'] = int(self.kfac_update_freq_base * factor)

@@ -1,0 +1,158 @@
+"""MPD K-FAC with explicit factor inversion ('inverse').
+
+CVPR-19-style model-parallel distributed preconditioning
+(reference: kfac/kfac_preconditioner_inv.py): factors are
+allreduce-averaged, each layer's inverse is computed by one owner rank
+(round-robin, rank_a == rank_g), and either the inverses or the
+owner-computed preconditioned gradients are broadcast back
+(``communicate_inverse_or_not``, default False -> broadcast pred).
+
+MI355X comm layout: the factor allreduce is ONE collective over a flat
+buffer; inverse / pred broadcasts are one flat bucket per owner rank,
+issued concurrently on rotating RCCL communicators.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List
+
+import torch
+import torch.nn as nn
+
+from kfac_pytorch_amd.ops.factors import ComputeA, ComputeG, factor_dims
+from kfac_pytorch_amd.ops.linalg import (add_diagonal_, inverse_precondition,
+                                         mat_inv)
+from kfac_pytorch_amd.preconditioner.base import KFACBase
+
+
+class KFACInverse(KFACBase):
+    """Model-parallel distributed K-FAC, explicit inverses
+    (reference class: kfac/kfac_preconditioner_inv.py:17)."""
+
+    def __init__(self, model, lr=0.1, damping=0.001, fac_update_freq=1,
+                 kfac_update_freq=1, communicate_inverse_or_not=False,
+                 kl_clip=0.001, factor_decay=0.95,
+                 exclude_vocabulary_size=None, hook_enabled=True,
+                 exclude_parts=''):
+        super().__init__(model=model, lr=lr, damping=damping,
+                         fac_update_freq=fac_update_freq,
+                         kfac_update_freq=kfac_update_freq,
+                         communicate_inverse_or_not=communicate_inverse_or_not,
+                         kl_clip=kl_clip, factor_decay=factor_decay,
+                         exclude_vocabulary_size=exclude_vocabulary_size,
+                         hook_enabled=hook_enabled,
+                         exclude_parts=exclude_parts)
+        self.computeA = ComputeA()
+        self.computeG = ComputeG()
+        self.m_inv_A: Dict[nn.Module, torch.Tensor] = {}
+        self.m_inv_G: Dict[nn.Module, torch.Tensor] = {}
+
+    # ------------------------------------------------------------- schedule
+    def schedule_module_ranks(self):
+        return self._round_robin_ranks(factor_wise=False)
+
+    # ---------------------------------------------------------------- state
+    def _init_state(self):
+        self._alloc_factor_buckets(owner_only=False)
+        if self.communicate_inverse_or_not:
+            self._alloc_inverse_buckets()
+        else:
+            self._alloc_pred_buckets()
+
+    def _alloc_inverse_buckets(self):
+        specs = []
+        for i, m in enumerate(self.modules):
+            rank_a, rank_g = self.module_ranks[m]
+            da, dg = factor_dims(m)
+            specs.append((f"invA{i}", (da, da), rank_a))
+            specs.append((f"invG{i}", (dg, dg), rank_g))
+        self.inv_buckets = self._alloc_owner_buckets(specs)
+        for i, m in enumerate(self.modules):
+            self.m_inv_A[m] = self._owner_view(self.inv_buckets, f"invA{i}")
+            self.m_inv_G[m] = self._owner_view(self.inv_buckets, f"invG{i}")
+
+    def _alloc_pred_buckets(self):
+        specs = []
+        for i, m in enumerate(self.modules):
+            rank_a, _ = self.module_ranks[m]
+            da, dg = factor_dims(m)
+            specs.append((f"pred{i}", (dg, da), rank_a))
+        self.pred_buckets = self._alloc_owner_buckets(specs)
+        for i, m in enumerate(self.modules):
+            self.m_precon_grad[m] = self._owner_view(
+                self.pred_buckets, f"pred{i}")
+        # inverses live only on the owner (never broadcast in this mode)
+        for m in self.modules:
+            da, dg = factor_dims(m)
+            dev = self._state_device()
+            rank_a, rank_g = self.module_ranks[m]
+            if self.comm.rank() == rank_a:
+                self.m_inv_A[m] = torch.zeros(da, da, device=dev)
+            if self.comm.rank() == rank_g:
+                self.m_inv_G[m] = torch.zeros(dg, dg, device=dev)
+
+    # -------------------------------------------------------------- factors
+    def _compute_factors(self):
+        """Local factors + running average, every rank, every layer
+        (reference :80-91); the running-average update is fused into the
+        factor kernel via out=/decay=."""
+        for m in self.modules:
+            self.computeA(self.m_a[m], m, out=self.m_A[m],
+                          decay=self.factor_decay)
+            self.computeG(self.m_g[m], m, batch_averaged=True,
+                          out=self.m_G[m], decay=self.factor_decay)
+
+    def _communicate_factors(self):
+        """ONE flat allreduce-average for all layers' A and G
+        (vs per-layer async bursts, reference :94-103)."""
+        self._allreduce_bucket_avg(self.factor_bucket)
+
+    # ------------------------------------------------------------- inverses
+    def _pi_damping(self, m) -> torch.Tensor:
+        """pi = sqrt((trA/dimA)/(trG/dimG)) (reference :121)."""
+        A, G = self.m_A[m], self.m_G[m]
+        trA = A.diagonal().sum() / A.shape[0]
+        trG = G.diagonal().sum() / G.shape[0]
+        return torch.sqrt(trA / trG)
+
+    def _compute_inverse(self):
+        """Owner rank inverts its layers' damped factors (reference
+        :109-129). The damped copy is temporary; m_A/m_G stay undamped
+        running averages."""
+        rank = self.comm.rank()
+        sqrt_damp = self.damping ** 0.5
+        for m in self.modules:
+            rank_a, rank_g = self.module_ranks[m]
+            if rank != rank_a and rank != rank_g:
+                continue
+            pi = self._pi_damping(m)
+            if rank == rank_a:
+                A = add_diagonal_(self.m_A[m].clone(), sqrt_damp * pi)
+                self.m_inv_A[m].copy_(mat_inv(A))
+            if rank == rank_g:
+                G = add_diagonal_(self.m_G[m].clone(), sqrt_damp / pi)
+                self.m_inv_G[m].copy_(mat_inv(G))
+
+    def _communicate_inverse(self):
+        self._broadcast_owner_buckets(self.inv_buckets)
+
+    # ----------------------------------------------------------------- pred
+    def _compute_pred(self):
+        if self.communicate_inverse_or_not:
+            # every rank preconditions every layer with broadcast inverses
+            for m in self.modules:
+                grad = self._get_grad(m)
+                self.m_precon_grad[m] = inverse_precondition(
+                    self.m_inv_A[m], self.m_inv_G[m], grad)
+        else:
+            # owner-only pred, broadcast after (reference default :41)
+            rank = self.comm.rank()
+            for m in self.modules:
+                rank_a, _ = self.module_ranks[m]
+                if rank == rank_a:
+                    grad = self._get_grad(m)
+                    self.m_precon_grad[m].copy_(inverse_precondition(
+                        self.m_inv_A[m], self.m_inv_G[m], grad))
+
+    def _communicate_pred(self):
+        self._broadcast_owner_buckets(self.pred_buckets)

@@ -1,0 +1,46 @@
+import os
+import socket
+import sys
+
+import pytest
+import torch
+import torch.distributed as dist
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+if REPO_ROOT not in sys.path:
+    sys.path.insert(0, REPO_ROOT)
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run via gpurun)")
+
+
+def free_port() -> int:
+    s = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.fixture
+def single_process_comm():
+    """torch.distributed gloo world_size=1 + kfac comm singleton."""
+    import kfac_pytorch_amd.parallel.comm as comm_mod
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    comm_mod.reset()
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{free_port()}",
+        world_size=1, rank=0)
+    comm_mod.init("Torch")
+    yield comm_mod.get_comm()
+    dist.destroy_process_group()
+    comm_mod.reset()
+
+
+@pytest.fixture
+def seeded():
+    torch.manual_seed(1234)
+    yield

@@ -1,0 +1,317 @@
+"""Multi-process (gloo, world_size=2) plumbing tests -- run on CPU.
+
+These cover the distributed paths the MI355X RCCL runs exercise:
+flat-bucket allreduce, rotating-group owner broadcasts, MPD factor
+averaging (exact parity with a single-rank full-batch run), and DP
+owner-only capture + pred broadcast consistency.
+"""
+
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+import torch.nn as nn
+import torch.nn.functional as F
+
+WORLD = 2
+
+
+def _init_worker(rank, world_size, tmpfile):
+    import kfac_pytorch_amd.parallel.comm as comm_mod
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpfile}",
+        world_size=world_size, rank=rank)
+    comm_mod.reset()
+    comm_mod.init("Torch")
+    return comm_mod.get_comm()
+
+
+def _run_spawn(fn, args=()):
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(fn, args=(WORLD, tmpfile) + args, nprocs=WORLD, join=True)
+
+
+class MLP(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.fc1 = nn.Linear(6, 8)
+        self.fc2 = nn.Linear(8, 8, bias=False)
+        self.fc3 = nn.Linear(8, 4)
+
+    def forward(self, x):
+        return self.fc3(F.relu(self.fc2(F.relu(self.fc1(x)))))
+
+
+def _global_batch(seed=11, n=8):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(n, 6, generator=g)
+    y = torch.randint(0, 4, (n,), generator=g)
+    return x, y
+
+
+def _train_grads(model, x, y):
+    model.zero_grad(set_to_none=False)
+    loss = F.cross_entropy(model(x), y)
+    loss.backward()
+
+
+# --------------------------------------------------------------------------
+def _worker_comm_primitives(rank, world, tmpfile):
+    comm = _init_worker(rank, world, tmpfile)
+    assert comm.size() == world and comm.rank() == rank
+
+    # allreduce average
+    t = torch.full((5,), float(rank + 1))
+    comm.allreduce(t, op=comm.Average)
+    torch.testing.assert_close(t, torch.full((5,), 1.5))
+
+    # async allreduce sum
+    t2 = torch.full((3,), float(rank))
+    h = comm.allreduce_async_(t2, op=comm.Sum)
+    comm.synchronize(h)
+    torch.testing.assert_close(t2, torch.full((3,), 1.0))
+
+    # broadcast from rank 1
+    t3 = torch.full((4,), float(rank * 7))
+    comm.broadcast(t3, src=1)
+    torch.testing.assert_close(t3, torch.full((4,), 7.0))
+
+    # rotating groups + concurrent owner broadcasts
+    n = comm.ensure_rotating_groups(2)
+    assert n == 2
+    a = torch.full((2,), float(rank))
+    b = torch.full((2,), float(rank + 10))
+    h1 = comm.broadcast_async_(a, src=0, group=comm.rotating_group(0))
+    h2 = comm.broadcast_async_(b, src=1, group=comm.rotating_group(1))
+    comm.synchronize([h1, h2])
+    torch.testing.assert_close(a, torch.full((2,), 0.0))
+    torch.testing.assert_close(b, torch.full((2,), 11.0))
+    dist.destroy_process_group()
+
+
+def test_comm_primitives():
+    _run_spawn(_worker_comm_primitives)
+
+
+# --------------------------------------------------------------------------
+def _worker_flat_bucket(rank, world, tmpfile):
+    from kfac_pytorch_amd.parallel.comm import FlatBucket
+    comm = _init_worker(rank, world, tmpfile)
+    b = FlatBucket(torch.float32)
+    b.add("x", torch.Size((3, 3)))
+    b.add("y", torch.Size((5,)))
+    b.freeze()
+    b.view("x").fill_(float(rank))
+    b.view("y").fill_(float(rank * 2))
+    comm.allreduce(b.buffer, op=comm.Average)
+    torch.testing.assert_close(b.view("x"), torch.full((3, 3), 0.5))
+    torch.testing.assert_close(b.view("y"), torch.full((5,), 1.0))
+    dist.destroy_process_group()
+
+
+def test_flat_bucket_allreduce():
+    _run_spawn(_worker_flat_bucket)
+
+
+# --------------------------------------------------------------------------
+def _worker_mpd_parity(rank, world, tmpfile, name):
+    """MPD K-FAC on 2 ranks with a split batch must exactly match a
+    1-rank run on the full batch: factor allreduce-average of half-batch
+    covariances == full-batch covariance, and grads are DDP-averaged."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+
+    torch.manual_seed(5)
+    model = MLP()
+    # broadcast initial params so both ranks agree
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+
+    KFAC = kfac.get_kfac_module(name)
+    pre = KFAC(model, lr=0.1, damping=0.01)
+
+    x, y = _global_batch()
+    half = x.size(0) // world
+    xs, ys = x[rank * half:(rank + 1) * half], y[rank * half:(rank + 1) * half]
+
+    for step in range(2):
+        _train_grads(model, xs, ys)
+        # emulate DDP gradient averaging
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        pre.step()
+
+    # single-rank full-batch simulation (world_size=1 semantics don't
+    # apply here: we recompute what the 2-rank MPD run should produce)
+    torch.manual_seed(5)
+    ref_model = MLP()
+    # ref_model has same init (same seed, but bcast was a no-op for rank0
+    # weights) -- verify
+    from kfac_pytorch_amd.ops.factors import ComputeA, ComputeG, \
+        update_running_avg
+    from kfac_pytorch_amd.ops.linalg import mat_eig, eigen_precondition, \
+        mat_inv, add_diagonal_, inverse_precondition
+    import math as _math
+
+    pre_ref = None  # manual reference below
+
+    # manual: run both half-batches, average factors, precondition
+    acts = {}
+    gouts = {}
+    handles = []
+    mods = [ref_model.fc1, ref_model.fc2, ref_model.fc3]
+
+    def fhook(mod, inp):
+        acts.setdefault(mod, []).append(inp[0].data)
+
+    def bhook(mod, gi, go):
+        gouts.setdefault(mod, []).append(go[0].data)
+
+    for m in mods:
+        handles.append(m.register_forward_pre_hook(fhook))
+        handles.append(m.register_full_backward_hook(bhook))
+
+    cA, cG = ComputeA(), ComputeG()
+    m_A = {m: torch.eye(m.in_features + (1 if m.bias is not None else 0))
+           for m in mods}
+    m_G = {m: torch.eye(m.out_features) for m in mods}
+    grads = None
+    for step in range(2):
+        acts.clear()
+        gouts.clear()
+        halves = []
+        for r in range(world):
+            xs_r = x[r * half:(r + 1) * half]
+            ys_r = y[r * half:(r + 1) * half]
+            _train_grads(ref_model, xs_r, ys_r)
+            halves.append([p.grad.clone() for p in ref_model.parameters()])
+        grads = [(a + b) / 2 for a, b in zip(*halves)]
+        for p, gr in zip(ref_model.parameters(), grads):
+            p.grad.data.copy_(gr)
+        # factors: average the two half-batch covariances
+        for m in mods:
+            A = sum(cA(a, m) for a in acts[m]) / world
+            G = sum(cG(g, m, True) for g in gouts[m]) / world
+            update_running_avg(A, m_A[m], 0.95)
+            update_running_avg(G, m_G[m], 0.95)
+        # precondition each module's grad
+        vg_sum = 0.0
+        vs = {}
+        for m in mods:
+            grad = m.weight.grad.data
+            if m.bias is not None:
+                grad = torch.cat([grad, m.bias.grad.data.view(-1, 1)], 1)
+            if name == "eigen":
+                dA, QA = mat_eig(m_A[m])
+                dG, QG = mat_eig(m_G[m])
+                dA = dA * (dA > 1e-10)
+                dG = dG * (dG > 1e-10)
+                v = eigen_precondition(QA, dA, QG, dG, grad, 0.01)
+            else:
+                A, G = m_A[m], m_G[m]
+                pi = torch.sqrt((A.trace() / A.shape[0]) /
+                                (G.trace() / G.shape[0]))
+                iA = mat_inv(add_diagonal_(A.clone(), (0.01 ** 0.5) * pi))
+                iG = mat_inv(add_diagonal_(G.clone(), (0.01 ** 0.5) / pi))
+                v = inverse_precondition(iA, iG, grad)
+            vs[m] = v
+            vg_sum += (v * grad * 0.1 ** 2).sum().item()
+        nu = min(1.0, _math.sqrt(0.001 / abs(vg_sum)))
+        for m in mods:
+            v = vs[m]
+            if m.bias is not None:
+                m.weight.grad.data.copy_(v[:, :-1] * nu)
+                m.bias.grad.data.copy_(v[:, -1] * nu)
+            else:
+                m.weight.grad.data.copy_(v * nu)
+
+    for p, q in zip(model.parameters(), ref_model.parameters()):
+        torch.testing.assert_close(p.grad, q.grad, rtol=1e-4, atol=1e-5)
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("name", ["eigen", "inverse"])
+def test_mpd_parity_with_full_batch(name):
+    _run_spawn(_worker_mpd_parity, args=(name,))
+
+
+# --------------------------------------------------------------------------
+def _worker_dp_consistency(rank, world, tmpfile, name):
+    """DP variants: owner-only capture, zero factor comm, pred broadcast
+    -> after step() every rank must hold identical preconditioned grads."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    torch.manual_seed(9)
+    model = MLP()
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+    pre = kfac.get_kfac_module(name)(model, damping=0.01)
+
+    # owner-gated hooks: each module's (a, g) saved only on its rank
+    x, y = _global_batch(seed=rank + 50)  # different local data per rank
+    for step in range(3):
+        _train_grads(model, x, y)
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        # check owner gating after the first backward
+        if step == 0:
+            for m, (ra, rg) in pre.module_ranks.items():
+                assert (m in pre.m_a) == (rank == ra)
+                assert (m in pre.m_g) == (rank == rg)
+        pre.step()
+        # every rank must now hold the same preconditioned grads
+        for p in model.parameters():
+            mine = p.grad.clone()
+            comm.broadcast(p.grad.data, src=0)
+            torch.testing.assert_close(mine, p.grad,
+                                       rtol=1e-5, atol=1e-6)
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("name", ["eigen_dp", "inverse_dp"])
+def test_dp_consistency_across_ranks(name):
+    _run_spawn(_worker_dp_consistency, args=(name,))
+
+
+# --------------------------------------------------------------------------
+def _worker_factor_wise(rank, world, tmpfile):
+    """MPD-eigen factor-wise distribution: rank_g = rank_a + 1 when
+    world > #modules (reference: kfac_preconditioner_eigen.py:66-94)."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    torch.manual_seed(3)
+
+    class One(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(4, 3)
+
+        def forward(self, x):
+            return self.fc(x)
+
+    model = One()
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+    pre = kfac.KFAC_EIGEN(model, damping=0.01)
+    x = torch.randn(4, 4)
+    model.zero_grad()
+    model(x).sum().backward()
+    for p in model.parameters():
+        comm.allreduce(p.grad.data, op=comm.Average)
+    pre.step()
+    (ra, rg), = pre.module_ranks.values()
+    assert (ra, rg) == (0, 1)  # factor-wise: world(2) > modules(1)
+    for p in model.parameters():
+        mine = p.grad.clone()
+        comm.broadcast(p.grad.data, src=0)
+        torch.testing.assert_close(mine, p.grad, rtol=1e-5, atol=1e-6)
+    dist.destroy_process_group()
+
+
+def test_factor_wise_distribution():
+    _run_spawn(_worker_factor_wise)
