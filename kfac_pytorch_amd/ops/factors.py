@@ -52,8 +52,16 @@ def extract_patches(x: torch.Tensor, kernel_size, stride, padding,
     Row-major patch ordering matches the flattened Conv2d weight layout
     ``[out, in*kh*kw]`` (in, kh, kw fastest-last), i.e. the same ordering
     the reference's double-unfold produces (kfac/utils.py:33-54).
+
+    GPU path: the HIP im2col kernel emits **bf16** patch rows directly
+    (the MFMA SYRK kernel's input dtype -- capture is bf16, accumulation
+    fp32, per the framework's GPU numerics policy). CPU path stays fp32.
     """
-    B = x.size(0)
+    if x.is_cuda:
+        from kfac_pytorch_amd.ops import _ext
+        return _ext.im2col(x.contiguous(), kernel_size[0], kernel_size[1],
+                           stride[0], stride[1], padding[0], padding[1],
+                           dilation[0], dilation[1])
     cols = F.unfold(x, kernel_size=kernel_size, stride=stride,
                     padding=padding, dilation=dilation)  # (B, C*kh*kw, L)
     return cols.transpose(1, 2).reshape(-1, cols.size(1))
@@ -77,6 +85,22 @@ def sym_factor(x: torch.Tensor, *, row_scale: float = 1.0, denom: float = 1.0,
     """
     if x.dim() != 2:
         raise ValueError(f"sym_factor expects 2-D input, got {tuple(x.shape)}")
+
+    if x.is_cuda:
+        # MI355X path: hand-written MFMA SYRK kernel, bf16 inputs / fp32
+        # accumulate, fused bias column + scale + running average.
+        # Raises loudly if the gfx950 extension is not built.
+        from kfac_pytorch_amd.ops import _ext
+        xb = x if x.dtype == torch.bfloat16 else x.bfloat16()
+        n = x.shape[1] + (1 if bias else 0)
+        if out is None:
+            out = torch.empty(n, n, device=x.device, dtype=torch.float32)
+            dec = -1.0
+        else:
+            dec = -1.0 if decay is None else float(decay)
+        return _ext.syrk_factor_(xb.contiguous(), out, float(row_scale),
+                                 float(denom), bool(bias), dec)
+
     x32 = x.float()
     rows, d = x32.shape
     n = d + 1 if bias else d
