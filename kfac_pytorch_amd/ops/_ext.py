@@ -82,8 +82,29 @@ def has_im2col() -> bool:
 
 
 # -- batched Jacobi symmetric eigensolver ------------------------------------
+def jacobi_eigh_batched(mats):
+    """Batched symmetric eigensolve of fp32 GPU matrices (one workgroup
+    each, all in one launch). Returns a list of (w, V) pairs; eigenvalues
+    UNSORTED (Jacobi diagonal order)."""
+    mod = _load()
+    W, V = mod.jacobi_eigh_batched(list(mats))
+    out = []
+    woff = moff = 0
+    for a in mats:
+        m = a.size(0)
+        out.append((W[woff:woff + m],
+                    V[moff:moff + m * m].view(m, m)))
+        woff += m
+        moff += m * m
+    return out
+
+
 def jacobi_eigh(x: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    return _load().jacobi_eigh(x)
+    """Single-matrix Jacobi eigensolve, eigenvalues sorted ascending
+    (torch.linalg.eigh contract)."""
+    (w, V), = jacobi_eigh_batched([x])
+    w, idx = torch.sort(w)
+    return w, V[:, idx].contiguous()
 
 
 def has_jacobi_eigh(m: int) -> bool:
@@ -91,4 +112,5 @@ def has_jacobi_eigh(m: int) -> bool:
         mod = _load()
     except RuntimeError:
         return False
-    return hasattr(mod, "jacobi_eigh") and m <= int(mod.jacobi_eigh_max_dim())
+    return (hasattr(mod, "jacobi_eigh_batched")
+            and m <= int(mod.jacobi_eigh_max_dim()))

@@ -367,6 +367,11 @@ torch::Tensor im2col_entry(torch::Tensor x, long kh, long kw, long sh,
   return out;
 }
 
+// defined in jacobi_eigh.hip
+int jacobi_eigh_max_dim();
+std::vector<torch::Tensor> jacobi_eigh_batched(
+    std::vector<torch::Tensor> mats);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native HIP kernels for distributed K-FAC";
   m.def("eigen_scale_", &eigen_scale_entry,
@@ -374,4 +379,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("syrk_factor_", &syrk_factor_entry,
         "fused bf16 MFMA factor product with bias column and running avg");
   m.def("im2col", &im2col_entry, "conv patch extraction to bf16 rows");
+  m.def("jacobi_eigh_batched", &jacobi_eigh_batched,
+        "batched LDS-resident Jacobi symmetric eigensolver (packed W, V)");
+  m.def("jacobi_eigh_max_dim", &jacobi_eigh_max_dim,
+        "largest dim the Jacobi kernel handles");
 }

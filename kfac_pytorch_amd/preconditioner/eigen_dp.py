@@ -17,10 +17,11 @@ import torch.nn as nn
 
 from kfac_pytorch_amd.ops.factors import factor_dims
 from kfac_pytorch_amd.ops.linalg import eigen_precondition, mat_eig
+from kfac_pytorch_amd.preconditioner.eigen import EigenComputeMixin
 from kfac_pytorch_amd.preconditioner.inverse_dp import KFACInverseDP
 
 
-class KFACEigenDP(KFACInverseDP):
+class KFACEigenDP(EigenComputeMixin, KFACInverseDP):
     """Distributed-preconditioning K-FAC, implicit eigen
     (reference class: kfac/kfac_preconditioner_eigen_dp.py:18)."""
 
@@ -58,19 +59,9 @@ class KFACEigenDP(KFACInverseDP):
 
     # ------------------------------------------------------------- inverses
     def _compute_inverse(self):
-        """Owner eigendecomposes its local factors, eigenvalues clamped
-        at eps (reference :62-75)."""
-        rank = self.comm.rank()
-        for m in self.modules:
-            rank_a, rank_g = self.module_ranks[m]
-            if rank == rank_a:
-                dA, QA = mat_eig(self.m_A[m])
-                self.m_QA[m].copy_(QA)
-                self.m_dA[m].copy_(dA * (dA > self.eps))
-            if rank == rank_g:
-                dG, QG = mat_eig(self.m_G[m])
-                self.m_QG[m].copy_(QG)
-                self.m_dG[m].copy_(dG * (dG > self.eps))
+        """Owner eigendecomposes its local factors (batched); eigenvalues
+        clamped at eps (reference :62-75)."""
+        self._eigendecompose_owned()
 
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):

@@ -105,7 +105,10 @@ def test_im2col_vs_cpu(ext, shape, k, s, p, dil):
 @requires_gpu
 def test_compute_factors_gpu_vs_cpu():
     """ComputeA/ComputeG full pipeline on GPU (HIP im2col + MFMA SYRK)
-    vs the fp32 CPU oracle path."""
+    vs the CPU oracle fed bf16-quantized inputs (isolates KERNEL
+    correctness from the intended bf16-capture quantization: on random
+    uncorrelated data the near-cancelling off-diagonal covariance sums
+    amplify input rounding, which is capture policy, not kernel error)."""
     import torch.nn as nn
     from kfac_pytorch_amd.ops.factors import ComputeA, ComputeG
     torch.manual_seed(7)
@@ -114,11 +117,15 @@ def test_compute_factors_gpu_vs_cpu():
     g = torch.randn(4, 16, 14, 14, device="cuda")
     A_gpu = ComputeA()(a, conv)
     G_gpu = ComputeG()(g, conv, True)
-    A_cpu = ComputeA()(a.cpu(), conv.cpu())
-    G_cpu = ComputeG()(g.cpu(), conv.cpu(), True)
-    torch.testing.assert_close(A_gpu.cpu(), A_cpu, rtol=3e-2, atol=3e-3)
-    torch.testing.assert_close(G_gpu.cpu(), G_cpu, rtol=3e-2,
-                               atol=3e-2)
+    a_q = a.cpu().bfloat16().float()
+    g_q = g.cpu().bfloat16().float()
+    A_cpu = ComputeA()(a_q, conv.cpu())
+    G_cpu = ComputeG()(g_q, conv.cpu(), True)
+    torch.testing.assert_close(A_gpu.cpu(), A_cpu, rtol=2e-3, atol=1e-4)
+    # G rows are scaled by B*spatial before the product; scale-relative
+    # tolerance
+    torch.testing.assert_close(G_gpu.cpu(), G_cpu, rtol=2e-3,
+                               atol=1e-2)
 
 
 @requires_gpu
@@ -170,3 +177,50 @@ def test_native_extension_is_loaded():
     assert _ext.available()
     from kfac_pytorch_amd.ops import _kfac_hip
     assert "_kfac_hip" in _kfac_hip.__file__
+
+
+@requires_gpu
+@pytest.mark.parametrize("m", [4, 27, 64, 65, 100, 128])
+def test_jacobi_eigh_vs_eigh(ext, m):
+    torch.manual_seed(m)
+    a = torch.randn(m, m)
+    a = (a @ a.t() / m + 0.1 * torch.eye(m)).cuda()
+    w, V = ext.jacobi_eigh(a)
+    w_ref = torch.linalg.eigvalsh(a)
+    torch.testing.assert_close(w, w_ref, rtol=1e-4, atol=1e-5)
+    # reconstruction + orthogonality (eigenvectors are sign/degenerate
+    # ambiguous, so compare via V diag(w) V^T)
+    recon = V @ torch.diag(w) @ V.t()
+    torch.testing.assert_close(recon, a, rtol=1e-4, atol=1e-4)
+    eye = torch.eye(m, device="cuda")
+    torch.testing.assert_close(V.t() @ V, eye, rtol=1e-4, atol=1e-4)
+
+
+@requires_gpu
+def test_jacobi_eigh_batched_mixed_sizes(ext):
+    torch.manual_seed(0)
+    mats = []
+    for m in (27, 65, 128, 10, 128):
+        a = torch.randn(m, m)
+        mats.append((a @ a.t() / m + 0.05 * torch.eye(m)).cuda())
+    results = ext.jacobi_eigh_batched(mats)
+    for a, (w, V) in zip(mats, results):
+        recon = V @ torch.diag(w) @ V.t()
+        torch.testing.assert_close(recon, a, rtol=1e-4, atol=1e-4)
+
+
+@requires_gpu
+def test_mat_eig_auto_uses_jacobi(ext):
+    from kfac_pytorch_amd.ops.linalg import mat_eig, mat_eig_multi
+    torch.manual_seed(2)
+    a = torch.randn(96, 96)
+    a = (a @ a.t() / 96 + 0.1 * torch.eye(96)).cuda()
+    d, Q = mat_eig(a, method="auto")
+    d_ref, _ = torch.linalg.eigh(a)
+    torch.testing.assert_close(d, d_ref, rtol=1e-4, atol=1e-5)
+    big = torch.randn(300, 300)
+    big = (big @ big.t() / 300 + 0.1 * torch.eye(300)).cuda()
+    outs = mat_eig_multi([a, big], need_sorted=False)
+    for src, (w, V) in zip([a, big], outs):
+        recon = V @ torch.diag(w) @ V.t()
+        torch.testing.assert_close(recon, src, rtol=1e-4, atol=1e-4)
