@@ -165,6 +165,13 @@ def main():
     vs_baseline = (images_per_sec / baseline_imgs
                    if args.kfac_name != "none" and bs == 32 else None)
 
+    if rank == 0 and precond is not None and \
+            hasattr(precond, "phase_times"):
+        per_step = {k: round(v / (args.steps + args.warmup) * 1000.0, 2)
+                    for k, v in precond.phase_times.items()}
+        print("KFAC_PHASES(ms/step):", json.dumps(per_step),
+              file=sys.stderr, flush=True)
+
     if rank == 0:
         result = {
             "metric": "images/sec ResNet-50 K-FAC",

@@ -336,6 +336,30 @@ class KFACBase(optim.Optimizer):
             torch._foreach_mul_(grads, nu)
 
     # ------------------------------------------------------------------ step
+    def _phase(self, name: str):
+        """Optional per-phase wall timing (KFAC_PHASE_TIMING=1): syncs the
+        GPU around each phase and accumulates seconds into
+        ``self.phase_times`` for perf attribution."""
+        import contextlib
+        import os
+        import time
+        if not os.environ.get("KFAC_PHASE_TIMING"):
+            return contextlib.nullcontext()
+
+        @contextlib.contextmanager
+        def timer():
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            yield
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            if not hasattr(self, "phase_times"):
+                self.phase_times = {}
+            self.phase_times[name] = self.phase_times.get(name, 0.0) + \
+                time.perf_counter() - t0
+        return timer()
+
     def step(self, closure=None, epoch=None):
         """One K-FAC step (4-phase template, reference
         kfac/kfac_preconditioner_base.py:185-230)."""
@@ -353,28 +377,35 @@ class KFACBase(optim.Optimizer):
 
         if self.steps % self.fac_update_freq == 0:
             if not self.exclude_compute_factor:
-                self._compute_factors()
+                with self._phase("compute_factor"):
+                    self._compute_factors()
             if not self.exclude_communicate_factor and self.comm.size() > 1:
-                self._communicate_factors()
+                with self._phase("comm_factor"):
+                    self._communicate_factors()
 
         if self.steps % self.kfac_update_freq == 0:
             if not self.exclude_compute_inverse:
-                self._compute_inverse()
+                with self._phase("compute_inverse"):
+                    self._compute_inverse()
             if (not self.exclude_communicate_inverse
                     and self.communicate_inverse_or_not
                     and self.comm.size() > 1):
-                self._communicate_inverse()
+                with self._phase("comm_inverse"):
+                    self._communicate_inverse()
 
         if not self.exclude_compute_inverse:
-            self._compute_pred()
+            with self._phase("compute_pred"):
+                self._compute_pred()
 
         if (not self.exclude_communicate_inverse
                 and not self.communicate_inverse_or_not
                 and self.comm.size() > 1):
-            self._communicate_pred()
+            with self._phase("comm_pred"):
+                self._communicate_pred()
 
         if not self.exclude_compute_inverse:
-            self._update_grad_in_place()
+            with self._phase("update_grad"):
+                self._update_grad_in_place()
 
         self.steps += 1
         self.m_a, self.m_g = {}, {}
