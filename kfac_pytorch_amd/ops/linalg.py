@@ -72,6 +72,16 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
     raise NotImplementedError(f"mat_eig method {method!r}")
 
 
+_EIG_STREAMS: list = []
+
+
+def _eig_streams(n: int = 8):
+    global _EIG_STREAMS
+    while len(_EIG_STREAMS) < n:
+        _EIG_STREAMS.append(torch.cuda.Stream())
+    return _EIG_STREAMS[:n]
+
+
 def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
     """Eigendecompose a list of symmetric matrices, batching every
     Jacobi-eligible GPU matrix into ONE kernel launch (the per-layer
@@ -98,10 +108,30 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                     w, idx = torch.sort(w)
                     V = V[:, idx].contiguous()
                 out[i] = (w, V)
-    for i, a in enumerate(mats):
-        if out[i] is None:
-            out[i] = mat_eig(a, method="eigh" if method != "jacobi"
-                             else "jacobi")
+
+    rest = [i for i in range(len(mats)) if out[i] is None]
+    if rest and mats[rest[0]].is_cuda and len(rest) > 1:
+        # stream-parallel the library eigensolves: each rocSOLVER syevd
+        # call runs at a few % GPU utilization (latency-bound internal
+        # iteration), so overlapping them across HIP streams recovers
+        # most of the serial-loop time.  Largest-first round-robin
+        # balances the streams.
+        ns = min(8, len(rest))
+        streams = _eig_streams(ns)
+        order = sorted(rest, key=lambda i: -mats[i].shape[-1])
+        cur = torch.cuda.current_stream()
+        for s in streams:
+            s.wait_stream(cur)
+        for k, i in enumerate(order):
+            with torch.cuda.stream(streams[k % ns]):
+                out[i] = mat_eig(mats[i], method="eigh")
+        for s in streams:
+            cur.wait_stream(s)
+        return out
+
+    for i in rest:
+        out[i] = mat_eig(mats[i], method="eigh" if method != "jacobi"
+                         else "jacobi")
     return out
 
 

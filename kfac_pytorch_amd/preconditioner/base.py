@@ -411,6 +411,62 @@ class KFACBase(optim.Optimizer):
         self.m_a, self.m_g = {}, {}
 
 
+def _iter_buckets(pre):
+    """All FlatBuckets a preconditioner allocated, keyed by name."""
+    out = {}
+    if getattr(pre, "factor_bucket", None) is not None:
+        out["factor"] = [pre.factor_bucket]
+    for attr in ("inv_buckets", "pred_buckets", "eig_buckets"):
+        if getattr(pre, attr, None) is not None:
+            out[attr] = list(getattr(pre, attr))
+    return out
+
+
+def kfac_state_dict(pre) -> dict:
+    """Serializable K-FAC state: step counter + every flat state buffer.
+
+    The reference never checkpoints K-FAC state (factors rebuild from
+    identity after restart -- SURVEY.md S5 'Checkpoint / resume'); this
+    makes preconditioning resume warm.  Valid to reload only with the
+    same model / world size / schedule.
+    """
+    state = {"steps": pre.steps, "buckets": {}}
+    for name, buckets in _iter_buckets(pre).items():
+        state["buckets"][name] = [
+            (b.buffer.detach().cpu() if b.buffer is not None else None)
+            for b in buckets]
+    return state
+
+
+def load_kfac_state_dict(pre, state: dict) -> None:
+    """Restore state saved by :func:`kfac_state_dict` (allocates the
+    buckets first if the preconditioner has not stepped yet)."""
+    if not pre._state_ready:
+        if pre.module_ranks is None:
+            pre.schedule_module_ranks()
+        pre._init_state()
+        pre._state_ready = True
+    pre.steps = int(state["steps"])
+    current = _iter_buckets(pre)
+    for name, buffers in state["buckets"].items():
+        if name not in current:
+            raise KeyError(f"checkpoint has bucket group {name!r} the "
+                           "preconditioner did not allocate (different "
+                           "algorithm or world size?)")
+        buckets = current[name]
+        if len(buckets) != len(buffers):
+            raise ValueError(f"bucket group {name!r}: checkpoint has "
+                             f"{len(buffers)} buffers, expected "
+                             f"{len(buckets)}")
+        for b, saved in zip(buckets, buffers):
+            if saved is None or b.buffer is None:
+                continue
+            if b.buffer.numel() != saved.numel():
+                raise ValueError(f"bucket group {name!r}: size mismatch "
+                                 "(different model/schedule?)")
+            b.buffer.copy_(saved.to(b.buffer.device))
+
+
 class KFACParamScheduler:
     """Epoch-schedule for damping and update frequencies
     (reference: kfac/kfac_preconditioner_base.py:233-301)."""
