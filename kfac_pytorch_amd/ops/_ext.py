@@ -44,6 +44,26 @@ def available() -> bool:
         return False
 
 
+_rccl_mod = None
+
+
+def load_rccl():
+    """Import the in-tree multi-stream RCCL communicator extension
+    (``_kfac_rccl``); raises RuntimeError if it is not built."""
+    global _rccl_mod
+    if _rccl_mod is None:
+        try:
+            from kfac_pytorch_amd.ops import _kfac_rccl  # built .so, in-tree
+            _rccl_mod = _kfac_rccl
+        except ImportError as e:  # pragma: no cover - GPU-box path
+            raise RuntimeError(
+                "kfac_pytorch_amd RCCL extension (_kfac_rccl) is not built. "
+                "Build in-tree with `python setup.py build_ext --inplace` "
+                f"(PYTORCH_ROCM_ARCH=gfx950). Original import error: {e}"
+            ) from e
+    return _rccl_mod
+
+
 # -- fused elementwise: V /= (dG dA^T + damping) ----------------------------
 def eigen_scale_(v: torch.Tensor, dG: torch.Tensor, dA: torch.Tensor,
                  damping: float) -> torch.Tensor:

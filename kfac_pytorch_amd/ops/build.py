@@ -18,30 +18,43 @@ CSRC = os.path.join(OPS_DIR, "csrc")
 EXT_NAME = "_kfac_hip"
 GFX_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
+# every extension in this package: name -> (source dir, extra link libs)
+EXTENSIONS = {
+    "_kfac_hip": ("csrc", []),
+    "_kfac_rccl": ("csrc_rccl", ["-lrccl"]),
+}
 
-def ext_path() -> str:
+
+def ext_path(name: str = EXT_NAME) -> str:
     suffix = importlib.machinery.EXTENSION_SUFFIXES[0]
-    return os.path.join(OPS_DIR, EXT_NAME + suffix)
+    return os.path.join(OPS_DIR, name + suffix)
 
 
-def _sources():
-    return [os.path.join(CSRC, f) for f in sorted(os.listdir(CSRC))
+def _sources(name: str = EXT_NAME):
+    src_dir = os.path.join(OPS_DIR, EXTENSIONS[name][0])
+    return [os.path.join(src_dir, f) for f in sorted(os.listdir(src_dir))
             if f.endswith(".hip")]
 
 
-def needs_build() -> bool:
-    so = ext_path()
+def needs_build(name: str = EXT_NAME) -> bool:
+    so = ext_path(name)
     if not os.path.exists(so):
         return True
     so_mtime = os.path.getmtime(so)
     return any(os.path.getmtime(s) > so_mtime
-               for s in _sources() + [os.path.abspath(__file__)])
+               for s in _sources(name) + [os.path.abspath(__file__)])
 
 
-def build(verbose: bool = True, force: bool = False) -> str:
-    """Compile the extension with hipcc for gfx950. Returns the .so path."""
-    so = ext_path()
-    if not force and not needs_build():
+def build_all(verbose: bool = True, force: bool = False):
+    """Compile every extension for gfx950. Returns the .so paths."""
+    return [build(verbose=verbose, force=force, name=n) for n in EXTENSIONS]
+
+
+def build(verbose: bool = True, force: bool = False,
+          name: str = EXT_NAME) -> str:
+    """Compile one extension with hipcc for gfx950. Returns the .so path."""
+    so = ext_path(name)
+    if not force and not needs_build(name):
         return so
 
     import torch
@@ -54,7 +67,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
         os.path.join(ce.ROCM_HOME or "/opt/rocm", "bin", "hipcc"),
         f"--offload-arch={GFX_ARCH}",
         "-O3", "-std=c++17", "-shared", "-fPIC",
-        f"-DTORCH_EXTENSION_NAME={EXT_NAME}",
+        f"-DTORCH_EXTENSION_NAME={name}",
         "-DTORCH_API_INCLUDE_EXTENSION_H",
         f"-D_GLIBCXX_USE_CXX11_ABI={int(torch._C._GLIBCXX_USE_CXX11_ABI)}",
     ]
@@ -62,11 +75,11 @@ def build(verbose: bool = True, force: bool = False) -> str:
     for inc in ce.include_paths("cuda"):
         cmd.append(f"-I{inc}")
     cmd.append(f"-I{py_include}")
-    cmd += _sources()
+    cmd += _sources(name)
     for lp in ce.library_paths("cuda"):
         cmd.append(f"-L{lp}")
     cmd += ["-ltorch", "-ltorch_hip", "-lc10", "-lc10_hip", "-ltorch_python",
-            "-lamdhip64", "-o", so]
+            "-lamdhip64"] + EXTENSIONS[name][1] + ["-o", so]
 
     if verbose:
         print(f"[kfac build] hipcc -> {os.path.basename(so)} "
@@ -77,9 +90,9 @@ def build(verbose: bool = True, force: bool = False) -> str:
         sys.stderr.write(e.stdout or "")
         sys.stderr.write(e.stderr or "")
         raise RuntimeError(
-            f"hipcc build of {EXT_NAME} failed (see output above)") from e
+            f"hipcc build of {name} failed (see output above)") from e
     return so
 
 
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
+    build_all(force="--force" in sys.argv)

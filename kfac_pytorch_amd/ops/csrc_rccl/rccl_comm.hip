@@ -1,0 +1,229 @@
+// MI355X-native multi-stream RCCL communicator for distributed K-FAC.
+//
+// Replaces the reference's raw-NCCL ``tcmm.Communicator``
+// (reference: packages/tcmm/src/communicator.cpp:5-117) with an
+// RCCL-over-xGMI design:
+//
+//  * N duplicate RCCL communicators, each pinned to its own non-blocking
+//    HIP stream, handed out round-robin per collective.  xGMI is
+//    point-to-point (7 links x ~153 GB/s per GPU) so independent
+//    broadcasts rooted at different owner ranks ride different links
+//    concurrently when issued on independent comms.
+//  * Bootstrap is rank-0 ``ncclGetUniqueId`` + exchange over the
+//    torch.distributed store (done by the Python wrapper in
+//    kfac_pytorch_amd/parallel/native.py) -- no MPI dependency
+//    (the reference bootstraps over MPI_Bcast, communicator.cpp:14-15).
+//  * Collectives are stream-ordered against the PyTorch current stream
+//    via HIP events in both directions (acquire before issue, release at
+//    join()), so callers never need a device-wide sync in the hot path
+//    (the reference host-syncs every op, tcmm_kernel.cu:107,154).
+//  * ``multi_bcast`` keeps the reference's fused compute-then-broadcast
+//    schedule (communicator.cpp:75-117): tensors >= min_tensor_size
+//    elements are assigned round-robin to owner ranks which run a Python
+//    callback (e.g. an eigendecomposition), then each output is broadcast
+//    from its owner on a rotating comm; small tensors are computed
+//    redundantly on every rank with no communication at all.
+//
+// Errors throw std::runtime_error (the reference print-and-exits,
+// communicator.h:15-42).
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <sstream>
+#include <string>
+#include <vector>
+
+#define HIPCHECK(cmd)                                                     \
+  do {                                                                    \
+    hipError_t e_ = (cmd);                                                \
+    if (e_ != hipSuccess) {                                               \
+      std::ostringstream oss;                                             \
+      oss << "HIP error " << hipGetErrorString(e_) << " at " << __FILE__  \
+          << ":" << __LINE__;                                             \
+      throw std::runtime_error(oss.str());                                \
+    }                                                                     \
+  } while (0)
+
+#define RCCLCHECK(cmd)                                                    \
+  do {                                                                    \
+    ncclResult_t r_ = (cmd);                                              \
+    if (r_ != ncclSuccess) {                                              \
+      std::ostringstream oss;                                             \
+      oss << "RCCL error " << ncclGetErrorString(r_) << " at "            \
+          << __FILE__ << ":" << __LINE__;                                 \
+      throw std::runtime_error(oss.str());                                \
+    }                                                                     \
+  } while (0)
+
+namespace {
+
+ncclDataType_t nccl_dtype(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return ncclFloat32;
+    case at::kDouble: return ncclFloat64;
+    case at::kHalf: return ncclFloat16;
+    case at::kBFloat16: return ncclBfloat16;
+    case at::kInt: return ncclInt32;
+    case at::kLong: return ncclInt64;
+    default:
+      throw std::runtime_error("unsupported dtype for RCCL collective: " +
+                               std::string(c10::toString(t.scalar_type())));
+  }
+}
+
+void check_tensor(const at::Tensor& t) {
+  TORCH_CHECK(t.is_cuda(), "RCCL collective needs a GPU tensor");
+  TORCH_CHECK(t.is_contiguous(), "RCCL collective needs a contiguous tensor");
+}
+
+}  // namespace
+
+class RcclCommunicator {
+ public:
+  RcclCommunicator(int rank, int size,
+                   const std::vector<std::string>& unique_ids)
+      : rank_(rank), size_(size), next_(0) {
+    TORCH_CHECK(!unique_ids.empty(), "need at least one RCCL unique id");
+    comms_.resize(unique_ids.size());
+    streams_.resize(unique_ids.size());
+    events_.resize(unique_ids.size());
+    for (size_t i = 0; i < unique_ids.size(); ++i) {
+      TORCH_CHECK(unique_ids[i].size() == sizeof(ncclUniqueId),
+                  "bad RCCL unique id size");
+      ncclUniqueId id;
+      std::memcpy(&id, unique_ids[i].data(), sizeof(id));
+      HIPCHECK(hipStreamCreateWithFlags(&streams_[i], hipStreamNonBlocking));
+      HIPCHECK(hipEventCreateWithFlags(&events_[i], hipEventDisableTiming));
+      RCCLCHECK(ncclCommInitRank(&comms_[i], size_, id, rank_));
+    }
+    HIPCHECK(hipEventCreateWithFlags(&acq_event_, hipEventDisableTiming));
+  }
+
+  ~RcclCommunicator() {
+    for (auto& c : comms_) ncclCommDestroy(c);
+    for (auto& s : streams_) hipStreamDestroy(s);
+    for (auto& e : events_) hipEventDestroy(e);
+    hipEventDestroy(acq_event_);
+  }
+
+  int rank() const { return rank_; }
+  int size() const { return size_; }
+  int num_comms() const { return static_cast<int>(comms_.size()); }
+
+  // In-place sum (or average) allreduce on the next rotating (comm, stream).
+  void all_reduce(at::Tensor t, bool average) {
+    check_tensor(t);
+    size_t slot = acquire(t);
+    RCCLCHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                            nccl_dtype(t), average ? ncclAvg : ncclSum,
+                            comms_[slot], streams_[slot]));
+  }
+
+  void reduce(at::Tensor t, int root, bool average) {
+    check_tensor(t);
+    size_t slot = acquire(t);
+    RCCLCHECK(ncclReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                         nccl_dtype(t), average ? ncclAvg : ncclSum, root,
+                         comms_[slot], streams_[slot]));
+  }
+
+  void broadcast(at::Tensor t, int root) {
+    check_tensor(t);
+    size_t slot = acquire(t);
+    RCCLCHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                            nccl_dtype(t), root, comms_[slot],
+                            streams_[slot]));
+  }
+
+  // Fused compute-then-broadcast (reference: communicator.cpp:75-117).
+  // ``op(input, output)`` is a Python callback run on the owner's torch
+  // current stream; its result is broadcast from the owner on a rotating
+  // comm.  Tensors smaller than ``min_numel`` are computed redundantly on
+  // EVERY rank (no communication) -- the latency of a ~64x64 broadcast on
+  // xGMI exceeds the redundant eigensolve cost.
+  void multi_bcast(std::vector<at::Tensor> tensors,
+                   std::vector<at::Tensor> outputs,
+                   const py::function& op, int64_t min_numel) {
+    TORCH_CHECK(tensors.size() == outputs.size(),
+                "multi_bcast: tensors/outputs length mismatch");
+    int owner = 0;
+    for (size_t i = 0; i < tensors.size(); ++i) {
+      if (tensors[i].numel() < min_numel) {
+        op(tensors[i], outputs[i]);  // redundant on all ranks, no comm
+        continue;
+      }
+      int root = owner % size_;
+      owner++;
+      if (root == rank_) {
+        op(tensors[i], outputs[i]);
+      }
+      broadcast(outputs[i], root);
+    }
+  }
+
+  // Host-blocking drain of all comm streams.
+  void synchronize() {
+    for (auto& s : streams_) HIPCHECK(hipStreamSynchronize(s));
+  }
+
+  // Stream-ordered drain: make the torch current stream wait on every comm
+  // stream (no host block).  Use this in the hot path.
+  void join() {
+    hipStream_t torch_stream = c10::hip::getCurrentHIPStream().stream();
+    for (size_t i = 0; i < streams_.size(); ++i) {
+      HIPCHECK(hipEventRecord(events_[i], streams_[i]));
+      HIPCHECK(hipStreamWaitEvent(torch_stream, events_[i], 0));
+    }
+  }
+
+ private:
+  // Order the collective after pending work on the torch current stream,
+  // return the rotating slot to issue on.
+  size_t acquire(const at::Tensor& t) {
+    size_t slot = next_++ % comms_.size();
+    hipStream_t torch_stream =
+        c10::hip::getCurrentHIPStream(t.get_device()).stream();
+    HIPCHECK(hipEventRecord(acq_event_, torch_stream));
+    HIPCHECK(hipStreamWaitEvent(streams_[slot], acq_event_, 0));
+    return slot;
+  }
+
+  int rank_, size_;
+  size_t next_;
+  std::vector<ncclComm_t> comms_;
+  std::vector<hipStream_t> streams_;
+  std::vector<hipEvent_t> events_;
+  hipEvent_t acq_event_;
+};
+
+static py::bytes get_unique_id() {
+  ncclUniqueId id;
+  RCCLCHECK(ncclGetUniqueId(&id));
+  return py::bytes(reinterpret_cast<const char*>(&id), sizeof(id));
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "multi-stream RCCL communicator (MI355X-native tcmm equivalent)";
+  m.def("get_unique_id", &get_unique_id,
+        "generate one RCCL unique id (call on rank 0, exchange via store)");
+  py::class_<RcclCommunicator>(m, "Communicator")
+      .def(py::init<int, int, const std::vector<std::string>&>(),
+           py::arg("rank"), py::arg("size"), py::arg("unique_ids"))
+      .def_property_readonly("rank", &RcclCommunicator::rank)
+      .def_property_readonly("size", &RcclCommunicator::size)
+      .def_property_readonly("num_comms", &RcclCommunicator::num_comms)
+      .def("all_reduce", &RcclCommunicator::all_reduce, py::arg("tensor"),
+           py::arg("average") = false)
+      .def("reduce", &RcclCommunicator::reduce, py::arg("tensor"),
+           py::arg("root"), py::arg("average") = false)
+      .def("broadcast", &RcclCommunicator::broadcast, py::arg("tensor"),
+           py::arg("root"))
+      .def("multi_bcast", &RcclCommunicator::multi_bcast, py::arg("tensors"),
+           py::arg("outputs"), py::arg("op"),
+           py::arg("min_numel") = 512 * 512)
+      .def("synchronize", &RcclCommunicator::synchronize)
+      .def("join", &RcclCommunicator::join);
+}

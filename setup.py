@@ -17,8 +17,8 @@ sys.path.insert(0, ROOT)
 
 class HipBuildExt(_build_ext):
     def run(self):
-        from kfac_pytorch_amd.ops.build import build
-        build(force=False)
+        from kfac_pytorch_amd.ops.build import build_all
+        build_all(force=False)
 
     def build_extensions(self):  # pragma: no cover
         pass
@@ -30,7 +30,8 @@ setup(
     description="MI355X-native distributed K-FAC for PyTorch-ROCm",
     packages=find_packages(include=["kfac_pytorch_amd",
                                     "kfac_pytorch_amd.*"]),
-    package_data={"kfac_pytorch_amd.ops": ["*.so", "csrc/*.hip"]},
+    package_data={"kfac_pytorch_amd.ops": ["*.so", "csrc/*.hip",
+                                           "csrc_rccl/*.hip"]},
     cmdclass={"build_ext": HipBuildExt},
     python_requires=">=3.9",
 )
