@@ -178,7 +178,18 @@ _MODELS = {
 
 
 def get_imagenet_model(name: str, num_classes: int = 1000) -> nn.Module:
+    """Name -> ImageNet model.  Covers the reference trainer's model list
+    (reference: examples/pytorch_imagenet_resnet.py:235-258): resnets +
+    resnext here, densenet/vgg/inception/mobilenet in imagenet_extras."""
     if name not in _MODELS:
+        from kfac_pytorch_amd.models import imagenet_extras as ex
+        extras = {
+            "densenet121": ex.densenet121, "densenet201": ex.densenet201,
+            "inceptionv3": ex.inception_v3, "inceptionv4": ex.inception_v4,
+            "mobilenetv2": ex.mobilenet_v2, "vgg16": ex.vgg16_imagenet,
+        }
+        if name in extras:
+            return extras[name](num_classes=num_classes)
         raise ValueError(f"unknown imagenet model {name!r}; "
-                         f"have {sorted(_MODELS)}")
+                         f"have {sorted(_MODELS) + sorted(extras)}")
     return _MODELS[name](num_classes=num_classes)

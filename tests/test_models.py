@@ -109,3 +109,29 @@ def test_seq_mean_keeps_factor_dims(single_process_comm, seeded):
         pre.step()
         assert pre.m_A[m.fc].shape == (9, 9)   # 8 + bias
         assert pre.m_G[m.fc].shape == (6, 6)
+
+
+def test_imagenet_extra_models_forward(seeded):
+    """The reference trainer's remaining model families
+    (examples/pytorch_imagenet_resnet.py:235-258) build and forward."""
+    for name, size in [("densenet121", 64), ("mobilenetv2", 64),
+                       ("inceptionv3", 128), ("inceptionv4", 128),
+                       ("vgg16", 64)]:
+        m = get_imagenet_model(name, num_classes=13)
+        y = m(torch.randn(1, 3, size, size))
+        assert y.shape == (1, 13), name
+
+
+def test_inceptionv4_kfac_hooks(single_process_comm, seeded):
+    """Inception-v4 is a BASELINE efficiency config (batch.sh:30);
+    K-FAC must hook every conv/linear and step through it."""
+    import kfac_pytorch_amd as kfac
+    m = get_imagenet_model("inceptionv4", num_classes=5)
+    pre = kfac.get_kfac_module("eigen_dp")(m, damping=0.003)
+    assert len(pre.modules) > 100
+    loss = F.cross_entropy(m(torch.randn(2, 3, 128, 128)),
+                           torch.tensor([0, 1]))
+    loss.backward()
+    pre.step()
+    assert all(torch.isfinite(p.grad).all() for p in m.parameters()
+               if p.grad is not None)
