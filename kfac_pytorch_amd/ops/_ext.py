@@ -44,6 +44,33 @@ def available() -> bool:
         return False
 
 
+_solver_mod = None
+
+
+def load_solver():
+    """Import the in-tree async/batched rocSOLVER extension
+    (``_kfac_solver``); raises RuntimeError if it is not built."""
+    global _solver_mod
+    if _solver_mod is None:
+        try:
+            from kfac_pytorch_amd.ops import _kfac_solver  # built .so
+            _solver_mod = _kfac_solver
+        except ImportError as e:  # pragma: no cover - GPU-box path
+            raise RuntimeError(
+                "kfac_pytorch_amd solver extension (_kfac_solver) is not "
+                "built. Build in-tree with `python setup.py build_ext "
+                f"--inplace`. Original import error: {e}"
+            ) from e
+    return _solver_mod
+
+
+def has_solver() -> bool:
+    try:
+        return load_solver() is not None
+    except RuntimeError:
+        return False
+
+
 _rccl_mod = None
 
 

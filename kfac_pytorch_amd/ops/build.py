@@ -22,6 +22,7 @@ GFX_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 EXTENSIONS = {
     "_kfac_hip": ("csrc", []),
     "_kfac_rccl": ("csrc_rccl", ["-lrccl"]),
+    "_kfac_solver": ("csrc_solver", ["-lrocsolver", "-lrocblas"]),
 }
 
 

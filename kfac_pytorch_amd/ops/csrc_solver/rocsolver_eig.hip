@@ -1,0 +1,200 @@
+// Async / batched symmetric eigensolves for K-FAC factors via rocSOLVER,
+// driven the MI355X way: device-resident arguments, persistent
+// handle+stream pool, NO host synchronization inside the hot path.
+//
+// Why this exists: torch.linalg.eigh's rocSOLVER wrapper host-syncs
+// every call (info check + workspace staging), so a ResNet-50 rank's
+// ~106 per-layer eigensolves serialize into ~1.7 s/step.  rocSOLVER
+// itself is stream-ordered with all-GPU arguments; issuing each solve
+// on a pool stream and joining once overlaps them, and factor dims
+// repeat heavily across a network, so same-dim factors additionally
+// batch into ONE ``syevdj_strided_batched`` call.
+//
+// This is the library tier of the eigensolver stack (the hand-written
+// LDS-Jacobi kernel in csrc/jacobi_eigh.hip covers m <= 128); it
+// replaces the reference's per-layer cuSOLVER ``cusolverDnSsyevd``
+// (reference: packages/tcmm/src/tcmm_kernel.cu:56-116 -- which even
+// cudaMalloc's its workspace and device-syncs per call).
+//
+// Layout note: rocSOLVER is column-major.  Inputs are symmetric, so no
+// transpose is needed going in; outputs leave eigenvectors in the
+// matrix buffer column-major, i.e. interpreting the same buffer
+// row-major, ROW i is the i-th eigenvector.  The Python wrapper
+// returns ``V.mT`` to restore the eigh contract (columns =
+// eigenvectors) without a copy.
+
+#include <hip/hip_runtime.h>
+#include <rocblas/rocblas.h>
+#include <rocsolver/rocsolver.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <sstream>
+#include <vector>
+
+#define HIPCHECK(cmd)                                                     \
+  do {                                                                    \
+    hipError_t e_ = (cmd);                                                \
+    TORCH_CHECK(e_ == hipSuccess, "HIP error ", hipGetErrorString(e_),    \
+                " at " __FILE__ ":", __LINE__);                           \
+  } while (0)
+
+#define ROCBLASCHECK(cmd)                                                 \
+  do {                                                                    \
+    rocblas_status s_ = (cmd);                                            \
+    TORCH_CHECK(s_ == rocblas_status_success, "rocSOLVER error ", (int)s_,\
+                " at " __FILE__ ":", __LINE__);                           \
+  } while (0)
+
+namespace {
+
+struct Slot {
+  rocblas_handle handle = nullptr;
+  hipStream_t stream = nullptr;
+  hipEvent_t event = nullptr;
+};
+
+constexpr int POOL = 8;
+
+Slot g_pool[POOL];
+rocblas_handle g_main_handle = nullptr;  // bound to torch current stream
+hipEvent_t g_acq_event = nullptr;
+bool g_init = false;
+
+void ensure_init() {
+  if (g_init) return;
+  for (int i = 0; i < POOL; ++i) {
+    ROCBLASCHECK(rocblas_create_handle(&g_pool[i].handle));
+    HIPCHECK(hipStreamCreateWithFlags(&g_pool[i].stream,
+                                      hipStreamNonBlocking));
+    HIPCHECK(hipEventCreateWithFlags(&g_pool[i].event,
+                                     hipEventDisableTiming));
+    ROCBLASCHECK(rocblas_set_stream(g_pool[i].handle, g_pool[i].stream));
+  }
+  ROCBLASCHECK(rocblas_create_handle(&g_main_handle));
+  HIPCHECK(hipEventCreateWithFlags(&g_acq_event, hipEventDisableTiming));
+  g_init = true;
+}
+
+}  // namespace
+
+// Batched eigensolve of same-size matrices: As (b, n, n) fp32 contiguous,
+// OVERWRITTEN with eigenvectors (column-major -> row i = eigenvector i
+// when read row-major).  Returns W (b, n) ascending and the info tensor
+// (b,) for deferred validation.  Runs on the torch current stream.
+std::vector<torch::Tensor> syevdj_batched_(torch::Tensor As) {
+  ensure_init();
+  TORCH_CHECK(As.is_cuda() && As.dim() == 3 && As.size(1) == As.size(2),
+              "syevdj_batched_: (b, n, n) GPU tensor required");
+  TORCH_CHECK(As.scalar_type() == at::kFloat && As.is_contiguous(),
+              "syevdj_batched_: fp32 contiguous required");
+  const long b = As.size(0);
+  const long n = As.size(1);
+  auto W = at::empty({b, n}, As.options());
+  auto info = at::empty({b}, As.options().dtype(at::kInt));
+  auto stream = c10::hip::getCurrentHIPStream();
+  ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  ROCBLASCHECK(rocsolver_ssyevdj_strided_batched(
+      g_main_handle, rocblas_evect_original, rocblas_fill_lower,
+      (rocblas_int)n, As.data_ptr<float>(), (rocblas_int)n,
+      (rocblas_stride)(n * n), W.data_ptr<float>(), (rocblas_stride)n,
+      info.data_ptr<int>(), (rocblas_int)b));
+  return {W, info};
+}
+
+// Pool-parallel eigensolves of differently-sized matrices: each matrix
+// is issued on one of POOL internal streams (round-robin) so the
+// host-async rocSOLVER calls overlap on-device; ``join`` makes the torch
+// current stream wait on every pool stream.  mats are OVERWRITTEN with
+// eigenvectors (same layout note as above).  Returns per-matrix W plus
+// one packed info tensor.
+std::vector<torch::Tensor> syevd_pool_(std::vector<torch::Tensor> mats) {
+  ensure_init();
+  TORCH_CHECK(!mats.empty(), "syevd_pool_: empty batch");
+  auto stream = c10::hip::getCurrentHIPStream();
+  // order pool streams after pending torch-stream work (factor updates)
+  HIPCHECK(hipEventRecord(g_acq_event, stream.stream()));
+  for (int i = 0; i < POOL; ++i) {
+    HIPCHECK(hipStreamWaitEvent(g_pool[i].stream, g_acq_event, 0));
+  }
+  auto opts = mats[0].options();
+  auto info = at::empty({(long)mats.size()}, opts.dtype(at::kInt));
+  std::vector<torch::Tensor> out;
+  out.reserve(mats.size() + 1);
+  std::vector<torch::Tensor> workE;
+  for (size_t i = 0; i < mats.size(); ++i) {
+    auto& A = mats[i];
+    TORCH_CHECK(A.is_cuda() && A.dim() == 2 && A.size(0) == A.size(1),
+                "syevd_pool_: square GPU matrices required");
+    TORCH_CHECK(A.scalar_type() == at::kFloat && A.is_contiguous(),
+                "syevd_pool_: fp32 contiguous required");
+    const long n = A.size(0);
+    auto W = at::empty({n}, opts);
+    auto E = at::empty({n}, opts);
+    workE.push_back(E);
+    Slot& s = g_pool[i % POOL];
+    ROCBLASCHECK(rocsolver_ssyevd(
+        s.handle, rocblas_evect_original, rocblas_fill_lower,
+        (rocblas_int)n, A.data_ptr<float>(), (rocblas_int)n,
+        W.data_ptr<float>(), E.data_ptr<float>(),
+        info.data_ptr<int>() + i));
+    out.push_back(W);
+  }
+  // join: torch stream waits on all pool streams
+  for (int i = 0; i < POOL && i < (int)mats.size(); ++i) {
+    HIPCHECK(hipEventRecord(g_pool[i].event, g_pool[i].stream));
+    HIPCHECK(hipStreamWaitEvent(stream.stream(), g_pool[i].event, 0));
+  }
+  out.push_back(info);
+  return out;
+}
+
+// Pool-parallel Cholesky inverse (potrf + potri) for the 'inverse'
+// K-FAC family: damped SPD factors, in-place, overlapped on the pool
+// (replaces the serial per-layer torch.cholesky_inverse loop;
+// reference: kfac/utils.py:11-20, kfac_preconditioner_inv.py:109-129).
+std::vector<torch::Tensor> potri_pool_(std::vector<torch::Tensor> mats) {
+  ensure_init();
+  TORCH_CHECK(!mats.empty(), "potri_pool_: empty batch");
+  auto stream = c10::hip::getCurrentHIPStream();
+  HIPCHECK(hipEventRecord(g_acq_event, stream.stream()));
+  for (int i = 0; i < POOL; ++i) {
+    HIPCHECK(hipStreamWaitEvent(g_pool[i].stream, g_acq_event, 0));
+  }
+  auto info = at::empty({(long)mats.size() * 2},
+                        mats[0].options().dtype(at::kInt));
+  for (size_t i = 0; i < mats.size(); ++i) {
+    auto& A = mats[i];
+    TORCH_CHECK(A.is_cuda() && A.dim() == 2 && A.size(0) == A.size(1) &&
+                    A.scalar_type() == at::kFloat && A.is_contiguous(),
+                "potri_pool_: square fp32 contiguous GPU matrices required");
+    const long n = A.size(0);
+    Slot& s = g_pool[i % POOL];
+    ROCBLASCHECK(rocsolver_spotrf(s.handle, rocblas_fill_lower,
+                                  (rocblas_int)n, A.data_ptr<float>(),
+                                  (rocblas_int)n,
+                                  info.data_ptr<int>() + 2 * i));
+    ROCBLASCHECK(rocsolver_spotri(s.handle, rocblas_fill_lower,
+                                  (rocblas_int)n, A.data_ptr<float>(),
+                                  (rocblas_int)n,
+                                  info.data_ptr<int>() + 2 * i + 1));
+  }
+  for (int i = 0; i < POOL && i < (int)mats.size(); ++i) {
+    HIPCHECK(hipEventRecord(g_pool[i].event, g_pool[i].stream));
+    HIPCHECK(hipStreamWaitEvent(stream.stream(), g_pool[i].event, 0));
+  }
+  return {info};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "async/batched rocSOLVER eigensolves for K-FAC (MI355X)";
+  m.def("syevdj_batched_", &syevdj_batched_,
+        "in-place batched symmetric eigensolve of (b,n,n); returns "
+        "(W, info); eigenvectors left row-major-transposed in input");
+  m.def("syevd_pool_", &syevd_pool_,
+        "in-place pool-stream-overlapped eigensolves of mixed sizes; "
+        "returns [W..., info]");
+  m.def("potri_pool_", &potri_pool_,
+        "in-place pool-stream-overlapped Cholesky inverse (lower) of "
+        "mixed sizes; returns [info]");
+}

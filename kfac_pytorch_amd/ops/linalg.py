@@ -72,34 +72,34 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
     raise NotImplementedError(f"mat_eig method {method!r}")
 
 
-_EIG_STREAMS: list = []
-
-
-def _eig_streams(n: int = 8):
-    global _EIG_STREAMS
-    while len(_EIG_STREAMS) < n:
-        _EIG_STREAMS.append(torch.cuda.Stream())
-    return _EIG_STREAMS[:n]
-
-
 def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
-    """Eigendecompose a list of symmetric matrices, batching every
-    Jacobi-eligible GPU matrix into ONE kernel launch (the per-layer
-    rocSOLVER loop the reference runs serializes ~50 eigensolves per
-    K-FAC step; here they run concurrently across CUs).
+    """Eigendecompose a list of symmetric matrices the MI355X way:
+
+    * every Jacobi-eligible matrix (m <= 128) goes into ONE hand-written
+      LDS-Jacobi kernel launch;
+    * remaining same-dim groups batch into ONE rocSOLVER
+      ``syevdj_strided_batched`` call (K-FAC factor dims repeat heavily
+      across a network);
+    * leftover singles are issued async on a persistent handle+stream
+      pool (``syevd_pool_``) and joined once.
+
+    torch.linalg.eigh host-syncs per call, so the per-layer loop the
+    reference runs serializes ~106 eigensolves per ResNet-50 step
+    (~1.7 s measured); this path removes every host sync.
 
     ``need_sorted=False`` skips the ascending-eigenvalue reorder --
     K-FAC's eigenvalue clamp and implicit-eigen preconditioner are
-    order-independent, so the hot path avoids ~2 launches per matrix.
-    Returns a list of (d, Q) aligned with ``mats``.
+    order-independent.  Returns a list of (d, Q) aligned with ``mats``;
+    Q may be a non-contiguous transposed view.
     """
     out = [None] * len(mats)
-    jac_idx = []
-    if method in ("auto", "jacobi") and len(mats) > 0 and mats[0].is_cuda:
+    if not mats:
+        return out
+    use_gpu = mats[0].is_cuda and method in ("auto", "jacobi")
+    if use_gpu:
         from kfac_pytorch_amd.ops import _ext
-        for i, a in enumerate(mats):
-            if a.is_cuda and _ext.has_jacobi_eigh(a.shape[-1]):
-                jac_idx.append(i)
+        jac_idx = [i for i, a in enumerate(mats)
+                   if _ext.has_jacobi_eigh(a.shape[-1])]
         if jac_idx:
             results = _ext.jacobi_eigh_batched(
                 [mats[i].contiguous() for i in jac_idx])
@@ -110,29 +110,74 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                 out[i] = (w, V)
 
     rest = [i for i in range(len(mats)) if out[i] is None]
-    if rest and mats[rest[0]].is_cuda and len(rest) > 1:
-        # stream-parallel the library eigensolves: each rocSOLVER syevd
-        # call runs at a few % GPU utilization (latency-bound internal
-        # iteration), so overlapping them across HIP streams recovers
-        # most of the serial-loop time.  Largest-first round-robin
-        # balances the streams.
-        ns = min(8, len(rest))
-        streams = _eig_streams(ns)
-        order = sorted(rest, key=lambda i: -mats[i].shape[-1])
-        cur = torch.cuda.current_stream()
-        for s in streams:
-            s.wait_stream(cur)
-        for k, i in enumerate(order):
-            with torch.cuda.stream(streams[k % ns]):
-                out[i] = mat_eig(mats[i], method="eigh")
-        for s in streams:
-            cur.wait_stream(s)
+    if rest and mats[rest[0]].is_cuda and method != "jacobi" \
+            and _solver_ok():
+        from kfac_pytorch_amd.ops import _ext
+        solver = _ext.load_solver()
+        # group same-dim matrices -> one batched syevdj call per group
+        groups = {}
+        for i in rest:
+            groups.setdefault(int(mats[i].shape[-1]), []).append(i)
+        singles = []
+        for n, idxs in sorted(groups.items()):
+            if len(idxs) < 2:
+                singles.extend(idxs)
+                continue
+            stacked = torch.stack([mats[i] for i in idxs]).contiguous()
+            W, _info = solver.syevdj_batched_(stacked)
+            for k, i in enumerate(idxs):
+                # rocSOLVER leaves eigenvectors column-major in the
+                # buffer: row-major row i = eigenvector i, so .mT gives
+                # the eigh contract (columns = eigenvectors)
+                out[i] = (W[k], stacked[k].mT)
+        if singles:
+            # largest-first keeps the 8 pool streams balanced
+            singles.sort(key=lambda i: -mats[i].shape[-1])
+            work = [mats[i].clone(
+                memory_format=torch.contiguous_format) for i in singles]
+            res = solver.syevd_pool_(work)
+            for k, i in enumerate(singles):
+                out[i] = (res[k], work[k].mT)
         return out
 
     for i in rest:
         out[i] = mat_eig(mats[i], method="eigh" if method != "jacobi"
                          else "jacobi")
     return out
+
+
+def _solver_ok() -> bool:
+    from kfac_pytorch_amd.ops import _ext
+    return _ext.has_solver()
+
+
+def mat_inv_multi(mats, damp_diag=None):
+    """Cholesky-invert a list of damped SPD GPU matrices with the
+    pool-stream rocSOLVER path (potrf+potri overlapped), falling back to
+    the serial torch path off-GPU.  ``damp_diag`` is an optional aligned
+    list of per-matrix diagonal damping values added BEFORE inversion
+    (the pi-damped copy the 'inverse' family makes,
+    reference: kfac/kfac_preconditioner_inv.py:109-129).
+
+    Returns a list of inverses (new tensors).
+    """
+    if not mats:
+        return []
+    work = []
+    for k, a in enumerate(mats):
+        w = a.clone(memory_format=torch.contiguous_format)
+        if damp_diag is not None:
+            w.diagonal().add_(damp_diag[k])
+        work.append(w)
+    if mats[0].is_cuda and _solver_ok():
+        from kfac_pytorch_amd.ops import _ext
+        solver = _ext.load_solver()
+        solver.potri_pool_(work)
+        # potri fills only the column-major lower triangle = our
+        # row-major UPPER triangle (the row-major lower still holds the
+        # damped input); mirror the upper triangle down
+        return [w.triu(0) + w.triu(1).mT for w in work]
+    return [mat_inv(w) for w in work]
 
 
 def eigen_precondition(QA: torch.Tensor, dA: torch.Tensor,

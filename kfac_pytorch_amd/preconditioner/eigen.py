@@ -27,20 +27,23 @@ from kfac_pytorch_amd.preconditioner.inverse import KFACInverse
 class EigenComputeMixin:
     """Shared owner-side eigendecomposition, MI355X-scheduled:
 
-    * small factors (<= the LDS-Jacobi cap) batch into ONE kernel launch;
-    * large factors use warm-started eigendecomposition TRACKING
-      (EigenTracker): rotate into the previous eigenbasis + a few block-
-      Jacobi rounds, with stream-parallel library cold starts -- the
-      factor is a 0.95-decay running average, so consecutive bases
-      differ by a small rotation and a full syevd per step is waste;
+    * small factors (<= the LDS-Jacobi cap) batch into ONE hand-written
+      kernel launch;
+    * same-dim factor groups batch into one rocSOLVER
+      ``syevdj_strided_batched`` call; leftover singles overlap on a
+      persistent async stream pool (no host syncs -- torch.linalg.eigh
+      would host-sync per matrix and serialize the whole set);
     * CPU falls back to plain eigh.
 
     Replaces the reference's serial per-layer eigh loop
     (reference: kfac_preconditioner_eigen.py:98-119).
-    """
 
-    #: set False (or env KFAC_EIG_TRACKER=0) to force full eigensolves
-    eig_tracker_enabled = True
+    ``KFAC_EIG_TRACKER=1`` additionally enables the experimental
+    perturbative eigenbasis tracker (ops/eig_tracker.py) for large
+    factors; off by default -- at factor_decay 0.95 the factor is 95%
+    fresh batch noise each step, which small-sample layers' bases do
+    not survive.
+    """
 
     def _eigendecompose_owned(self):
         import os
@@ -57,8 +60,7 @@ class EigenComputeMixin:
         mats = [self.m_A[mod] if kind == "A" else self.m_G[mod]
                 for mod, kind in work]
 
-        use_tracker = (self.eig_tracker_enabled
-                       and os.environ.get("KFAC_EIG_TRACKER", "1") != "0"
+        use_tracker = (os.environ.get("KFAC_EIG_TRACKER", "0") == "1"
                        and mats[0].is_cuda)
         if use_tracker:
             from kfac_pytorch_amd.ops import _ext

