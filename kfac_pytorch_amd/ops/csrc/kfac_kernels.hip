@@ -55,6 +55,24 @@ __global__ void eigen_scale_kernel(float* __restrict__ v,
   }
 }
 
+// batched: v[b, ng, na] /= (dG[b, ng] dA[b, na]^T + damping)
+__global__ void eigen_scale_batched_kernel(float* __restrict__ v,
+                                           const float* __restrict__ dG,
+                                           const float* __restrict__ dA,
+                                           float damping, long nb, long ng,
+                                           long na) {
+  const long per = ng * na;
+  const long total = nb * per;
+  for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * blockDim.x) {
+    const long b = idx / per;
+    const long r = idx - b * per;
+    const long i = r / na;
+    const long j = r - i * na;
+    v[idx] /= (dG[b * ng + i] * dA[b * na + j] + damping);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // SYRK factor kernel: Ctmp += [X|1]^T [X|1]  (raw accumulation; the
 // scale s^2/denom and the running average land in the epilogue).
@@ -286,6 +304,28 @@ void eigen_scale_entry(torch::Tensor v, torch::Tensor dG, torch::Tensor dA,
   HIP_CHECK(hipGetLastError());
 }
 
+void eigen_scale_batched_entry(torch::Tensor v, torch::Tensor dG,
+                               torch::Tensor dA, double damping) {
+  TORCH_CHECK(v.is_cuda() && v.dim() == 3 && v.is_contiguous() &&
+                  dG.is_contiguous() && dA.is_contiguous(),
+              "eigen_scale_batched_: contiguous GPU tensors required");
+  TORCH_CHECK(v.scalar_type() == at::kFloat &&
+                  dG.scalar_type() == at::kFloat &&
+                  dA.scalar_type() == at::kFloat,
+              "eigen_scale_batched_: fp32 only");
+  const long nb = v.size(0), ng = v.size(1), na = v.size(2);
+  TORCH_CHECK(dG.numel() == nb * ng && dA.numel() == nb * na,
+              "eigen_scale_batched_: shape mismatch");
+  auto stream = c10::hip::getCurrentHIPStream();
+  const long total = nb * ng * na;
+  const int block = 256;
+  const int grid = (int)std::min<long>((total + block - 1) / block, 4096);
+  eigen_scale_batched_kernel<<<grid, block, 0, stream.stream()>>>(
+      v.data_ptr<float>(), dG.data_ptr<float>(), dA.data_ptr<float>(),
+      (float)damping, nb, ng, na);
+  HIP_CHECK(hipGetLastError());
+}
+
 torch::Tensor syrk_factor_entry(torch::Tensor x, torch::Tensor out,
                                 double row_scale, double denom, bool bias,
                                 double decay) {
@@ -377,6 +417,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native HIP kernels for distributed K-FAC";
   m.def("eigen_scale_", &eigen_scale_entry,
         "in-place V /= (dG dA^T + damping)");
+  m.def("eigen_scale_batched_", &eigen_scale_batched_entry,
+        "in-place batched V[b] /= (dG[b] dA[b]^T + damping)");
   m.def("syrk_factor_", &syrk_factor_entry,
         "fused bf16 MFMA factor product with bias column and running avg");
   m.def("im2col", &im2col_entry, "conv patch extraction to bf16 rows");

@@ -19,8 +19,8 @@ import torch
 import torch.nn as nn
 
 from kfac_pytorch_amd.ops.factors import factor_dims
-from kfac_pytorch_amd.ops.linalg import (eigen_precondition, mat_eig,
-                                         mat_eig_multi)
+from kfac_pytorch_amd.ops.linalg import (eigen_precondition_multi,
+                                         mat_eig, mat_eig_multi)
 from kfac_pytorch_amd.preconditioner.inverse import KFACInverse
 
 
@@ -162,9 +162,12 @@ class KFACEigen(EigenComputeMixin, KFACInverse):
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):
         """Implicit-eigen preconditioning on every rank (reference
-        :137-144)."""
-        for m in self.modules:
-            grad = self._get_grad(m)
-            self.m_precon_grad[m] = eigen_precondition(
-                self.m_QA[m], self.m_dA[m], self.m_QG[m], self.m_dG[m],
-                grad, self.damping)
+        :137-144), same-shape layers batched."""
+        preds = eigen_precondition_multi(
+            [self.m_QA[m] for m in self.modules],
+            [self.m_dA[m] for m in self.modules],
+            [self.m_QG[m] for m in self.modules],
+            [self.m_dG[m] for m in self.modules],
+            [self._get_grad(m) for m in self.modules], self.damping)
+        for m, p in zip(self.modules, preds):
+            self.m_precon_grad[m] = p

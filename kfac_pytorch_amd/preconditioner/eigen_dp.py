@@ -16,7 +16,8 @@ import torch
 import torch.nn as nn
 
 from kfac_pytorch_amd.ops.factors import factor_dims
-from kfac_pytorch_amd.ops.linalg import eigen_precondition, mat_eig
+from kfac_pytorch_amd.ops.linalg import (eigen_precondition_multi,
+                                         mat_eig)
 from kfac_pytorch_amd.preconditioner.eigen import EigenComputeMixin
 from kfac_pytorch_amd.preconditioner.inverse_dp import KFACInverseDP
 
@@ -66,14 +67,17 @@ class KFACEigenDP(EigenComputeMixin, KFACInverseDP):
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):
         """Owner-only implicit-eigen preconditioning of the averaged
-        gradient (reference :78-93)."""
+        gradient (reference :78-93), same-shape layers batched
+        (eigen_precondition_multi)."""
         assert not self.communicate_inverse_or_not
         rank = self.comm.rank()
-        for m in self.modules:
-            rank_a, rank_g = self.module_ranks[m]
-            assert rank_a == rank_g
-            if rank == rank_a:
-                grad = self._get_grad(m)
-                self.m_precon_grad[m].copy_(eigen_precondition(
-                    self.m_QA[m], self.m_dA[m], self.m_QG[m], self.m_dG[m],
-                    grad, self.damping))
+        owned = [m for m in self.modules
+                 if rank == self.module_ranks[m][0]]
+        if not owned:
+            return
+        preds = eigen_precondition_multi(
+            [self.m_QA[m] for m in owned], [self.m_dA[m] for m in owned],
+            [self.m_QG[m] for m in owned], [self.m_dG[m] for m in owned],
+            [self._get_grad(m) for m in owned], self.damping)
+        for m, p in zip(owned, preds):
+            self.m_precon_grad[m].copy_(p)
