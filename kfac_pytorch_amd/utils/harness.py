@@ -165,3 +165,43 @@ class PhaseTimers:
         total = sum(s.values())
         parts = " ".join(f"{p}={t * 1000:.1f}ms" for p, t in s.items())
         return f"iter={total * 1000:.1f}ms [{parts}]"
+
+
+class MultiEpochsDataLoader(torch.utils.data.DataLoader):
+    """DataLoader whose worker pool and iterator persist across epochs
+    (capability analog of the reference's MultiEpochsDataLoader,
+    examples/utils.py:93-121 -- avoids worker restart cost per epoch)."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._DataLoader__initialized = False
+        self.batch_sampler = _RepeatSampler(self.batch_sampler)
+        self._DataLoader__initialized = True
+        self.iterator = super().__iter__()
+
+    def __len__(self):
+        return len(self.batch_sampler.sampler)
+
+    def __iter__(self):
+        for _ in range(len(self)):
+            yield next(self.iterator)
+
+
+class _RepeatSampler:
+    """Endlessly repeating batch sampler (examples/utils.py:106-121)."""
+
+    def __init__(self, sampler):
+        self.sampler = sampler
+
+    def __iter__(self):
+        while True:
+            yield from iter(self.sampler)
+
+
+def generate_pseudo_labels(output: torch.Tensor) -> torch.Tensor:
+    """Sample labels from the model's predictive distribution -- the
+    Monte-Carlo Fisher ('F1mc') option (reference:
+    examples/utils.py:83-90; --kfac-type arg at
+    examples/pytorch_cifar10_resnet.py:74-75)."""
+    probs = torch.softmax(output.detach(), dim=-1)
+    return torch.multinomial(probs, 1).squeeze(-1)
