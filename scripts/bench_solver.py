@@ -51,8 +51,12 @@ def main():
     solver = _ext.load_solver()
     b = args.batch
 
-    has_magma = torch.has_magma
-    print(f"torch.has_magma = {has_magma}")
+    try:
+        has_magma = torch.backends.cuda.is_built() and \
+            torch._C._has_magma
+    except AttributeError:
+        has_magma = False
+    print(f"magma available = {has_magma}")
 
     hdr = (f"{'m':>6} {'eigh':>9} {'magma':>9} {'pool/8':>9} "
            f"{'syevdj/8':>10} {'jac/8':>9} {'potri/8':>9} {'chol_inv':>9}")
