@@ -224,3 +224,37 @@ def test_mat_eig_auto_uses_jacobi(ext):
     for src, (w, V) in zip([a, big], outs):
         recon = V @ torch.diag(w) @ V.t()
         torch.testing.assert_close(recon, src, rtol=1e-4, atol=1e-4)
+
+
+def test_eigen_scale_batched(ext):
+    torch.manual_seed(5)
+    nb, ng, na = 4, 33, 57
+    v = torch.randn(nb, ng, na, device="cuda")
+    dG = torch.rand(nb, ng, device="cuda")
+    dA = torch.rand(nb, na, device="cuda")
+    ref = v / (dG.unsqueeze(2) * dA.unsqueeze(1) + 0.003)
+    ext.eigen_scale_batched_(v, dG, dA, 0.003)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(v, ref, rtol=1e-6, atol=1e-7)
+
+
+def test_eigen_precondition_multi_matches_single():
+    """Shape-grouped bmm path vs the per-layer oracle."""
+    from kfac_pytorch_amd.ops.linalg import (eigen_precondition,
+                                             eigen_precondition_multi)
+    torch.manual_seed(6)
+    shapes = [(64, 129), (64, 129), (64, 129), (128, 257), (32, 65)]
+    QAs, dAs, QGs, dGs, grads = [], [], [], [], []
+    for dg, da in shapes:
+        QAs.append(torch.linalg.qr(
+            torch.randn(da, da, device="cuda"))[0].contiguous())
+        QGs.append(torch.linalg.qr(
+            torch.randn(dg, dg, device="cuda"))[0].contiguous())
+        dAs.append(torch.rand(da, device="cuda"))
+        dGs.append(torch.rand(dg, device="cuda"))
+        grads.append(torch.randn(dg, da, device="cuda"))
+    outs = eigen_precondition_multi(QAs, dAs, QGs, dGs, grads, 0.002)
+    for i in range(len(shapes)):
+        ref = eigen_precondition(QAs[i], dAs[i], QGs[i], dGs[i],
+                                 grads[i].clone(), 0.002)
+        torch.testing.assert_close(outs[i], ref, rtol=1e-4, atol=1e-5)
