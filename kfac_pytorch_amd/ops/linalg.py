@@ -72,6 +72,13 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
     raise NotImplementedError(f"mat_eig method {method!r}")
 
 
+# dispatch cutoffs, measured on MI355X (profiles/bench_solver.log):
+# hand-written LDS-Jacobi wins at m <= 64 (0.33 ms/matrix batched);
+# rocSOLVER syevdj_strided_batched wins 65..1152; async-pool syevd above.
+JAC_DISPATCH_MAX = 64
+SYEVDJ_MAX = 1152
+
+
 def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
     """Eigendecompose a list of symmetric matrices the MI355X way:
 
@@ -99,7 +106,8 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
     if use_gpu:
         from kfac_pytorch_amd.ops import _ext
         jac_idx = [i for i, a in enumerate(mats)
-                   if _ext.has_jacobi_eigh(a.shape[-1])]
+                   if a.shape[-1] <= JAC_DISPATCH_MAX
+                   and _ext.has_jacobi_eigh(a.shape[-1])]
         if jac_idx:
             results = _ext.jacobi_eigh_batched(
                 [mats[i].contiguous() for i in jac_idx])
@@ -115,12 +123,15 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
         from kfac_pytorch_amd.ops import _ext
         solver = _ext.load_solver()
         # group same-dim matrices -> one batched syevdj call per group
+        # (measured on MI355X: syevdj_strided_batched wins up to
+        # m~1152, collapses beyond -- 1.45 s at m=4608; big matrices go
+        # to the async syevd pool, profiles/bench_solver)
         groups = {}
         for i in rest:
             groups.setdefault(int(mats[i].shape[-1]), []).append(i)
         singles = []
         for n, idxs in sorted(groups.items()):
-            if len(idxs) < 2:
+            if len(idxs) < 2 or n > SYEVDJ_MAX:
                 singles.extend(idxs)
                 continue
             stacked = torch.stack([mats[i] for i in idxs]).contiguous()

@@ -38,12 +38,19 @@ class EigenComputeMixin:
     Replaces the reference's serial per-layer eigh loop
     (reference: kfac_preconditioner_eigen.py:98-119).
 
-    ``KFAC_EIG_TRACKER=1`` additionally enables the experimental
-    perturbative eigenbasis tracker (ops/eig_tracker.py) for large
-    factors; off by default -- at factor_decay 0.95 the factor is 95%
-    fresh batch noise each step, which small-sample layers' bases do
-    not survive.
+    Factors larger than ``TRACK_MIN`` additionally go through the
+    health-gated perturbative eigenbasis tracker (ops/eig_tracker.py):
+    when the factor drifts slowly between updates the full library
+    solve (118 ms at m=4608) is replaced by ~7 GEMMs; the per-factor
+    health gate falls back to the exact solve whenever first-order
+    tracking is invalid (e.g. heavy batch noise), so results stay
+    within the gate tolerance of the exact decomposition always.
+    ``KFAC_EIG_TRACKER=0`` disables it.
     """
+
+    #: factors above this dim use the perturbative tracker (the
+    #: syevdj-batched tier below it is already ~ms-cheap)
+    TRACK_MIN = 768
 
     def _eigendecompose_owned(self):
         import os
@@ -60,16 +67,15 @@ class EigenComputeMixin:
         mats = [self.m_A[mod] if kind == "A" else self.m_G[mod]
                 for mod, kind in work]
 
-        use_tracker = (os.environ.get("KFAC_EIG_TRACKER", "0") == "1"
+        use_tracker = (os.environ.get("KFAC_EIG_TRACKER", "1") != "0"
                        and mats[0].is_cuda)
         if use_tracker:
-            from kfac_pytorch_amd.ops import _ext
             from kfac_pytorch_amd.ops.eig_tracker import (EigenTracker,
                                                           tracked_eig_multi)
             if not hasattr(self, "_eig_trackers"):
                 self._eig_trackers = {}
             small = [i for i, a in enumerate(mats)
-                     if _ext.has_jacobi_eigh(a.shape[-1])]
+                     if a.shape[-1] <= self.TRACK_MIN]
             big = [i for i in range(len(mats)) if i not in set(small)]
             results = [None] * len(mats)
             if small:
