@@ -81,9 +81,12 @@ class KFACInverseDP(KFACInverse):
     # ------------------------------------------------------------- inverses
     def _compute_inverse(self):
         """Owner inverts its own locally-built factors (reference
-        :98-123)."""
+        :98-123), all of them issued as one pool-overlapped potrf+potri
+        batch on GPU (mat_inv_multi)."""
+        from kfac_pytorch_amd.ops.linalg import mat_inv_multi
         rank = self.comm.rank()
         sqrt_damp = self.damping ** 0.5
+        mats, damps, dests = [], [], []
         for m in self.modules:
             rank_a, rank_g = self.module_ranks[m]
             if rank != rank_a and rank != rank_g:
@@ -93,11 +96,15 @@ class KFACInverseDP(KFACInverse):
             else:
                 pi = 1.0
             if rank == rank_a:
-                A = add_diagonal_(self.m_A[m].clone(), sqrt_damp * pi)
-                self.m_inv_A[m].copy_(mat_inv(A))
+                mats.append(self.m_A[m])
+                damps.append(sqrt_damp * pi)
+                dests.append(self.m_inv_A[m])
             if rank == rank_g:
-                G = add_diagonal_(self.m_G[m].clone(), sqrt_damp / pi)
-                self.m_inv_G[m].copy_(mat_inv(G))
+                mats.append(self.m_G[m])
+                damps.append(sqrt_damp / pi)
+                dests.append(self.m_inv_G[m])
+        for inv, dst in zip(mat_inv_multi(mats, damp_diag=damps), dests):
+            dst.copy_(inv)
 
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):
