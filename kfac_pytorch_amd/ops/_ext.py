@@ -97,6 +97,12 @@ def eigen_scale_(v: torch.Tensor, dG: torch.Tensor, dA: torch.Tensor,
     return _load().eigen_scale_(v, dG, dA, float(damping))
 
 
+def eigen_scale_batched_(v: torch.Tensor, dG: torch.Tensor,
+                         dA: torch.Tensor, damping: float) -> None:
+    """In-place batched V[b] /= (dG[b] dA[b]^T + damping)."""
+    _load().eigen_scale_batched_(v, dG, dA, float(damping))
+
+
 # -- MFMA SYRK factor kernel -------------------------------------------------
 def syrk_factor_(x: torch.Tensor, out: torch.Tensor, row_scale: float,
                  denom: float, bias: bool, decay: float) -> torch.Tensor:
