@@ -31,6 +31,7 @@
 #include <rccl/rccl.h>
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPCachingAllocator.h>
 
 #include <sstream>
 #include <string>
@@ -181,13 +182,18 @@ class RcclCommunicator {
 
  private:
   // Order the collective after pending work on the torch current stream,
-  // return the rotating slot to issue on.
+  // return the rotating slot to issue on.  recordStream keeps the
+  // caching allocator from re-using the tensor's memory while the
+  // collective is still in flight on our stream.
   size_t acquire(const at::Tensor& t) {
     size_t slot = next_++ % comms_.size();
     hipStream_t torch_stream =
         c10::hip::getCurrentHIPStream(t.get_device()).stream();
     HIPCHECK(hipEventRecord(acq_event_, torch_stream));
     HIPCHECK(hipStreamWaitEvent(streams_[slot], acq_event_, 0));
+    c10::hip::HIPCachingAllocator::recordStream(
+        t.storage().data_ptr(),
+        c10::hip::getStreamFromExternal(streams_[slot], t.get_device()));
     return slot;
   }
 

@@ -28,9 +28,19 @@
 #include <rocsolver/rocsolver.h>
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
+#include <c10/hip/HIPCachingAllocator.h>
 
 #include <sstream>
 #include <vector>
+
+// Tell the caching allocator a tensor is in use on one of our pool
+// streams, so freeing it on the torch stream cannot hand its memory to
+// another op while the async rocSOLVER call is still running.
+static void record_on(const at::Tensor& t, hipStream_t s) {
+  c10::hip::HIPCachingAllocator::recordStream(
+      t.storage().data_ptr(),
+      c10::hip::getStreamFromExternal(s, t.get_device()));
+}
 
 #define HIPCHECK(cmd)                                                     \
   do {                                                                    \
@@ -133,6 +143,10 @@ std::vector<torch::Tensor> syevd_pool_(std::vector<torch::Tensor> mats) {
     auto E = at::empty({n}, opts);
     workE.push_back(E);
     Slot& s = g_pool[i % POOL];
+    record_on(A, s.stream);
+    record_on(W, s.stream);
+    record_on(E, s.stream);
+    record_on(info, s.stream);
     ROCBLASCHECK(rocsolver_ssyevd(
         s.handle, rocblas_evect_original, rocblas_fill_lower,
         (rocblas_int)n, A.data_ptr<float>(), (rocblas_int)n,
@@ -170,6 +184,8 @@ std::vector<torch::Tensor> potri_pool_(std::vector<torch::Tensor> mats) {
                 "potri_pool_: square fp32 contiguous GPU matrices required");
     const long n = A.size(0);
     Slot& s = g_pool[i % POOL];
+    record_on(A, s.stream);
+    record_on(info, s.stream);
     ROCBLASCHECK(rocsolver_spotrf(s.handle, rocblas_fill_lower,
                                   (rocblas_int)n, A.data_ptr<float>(),
                                   (rocblas_int)n,
