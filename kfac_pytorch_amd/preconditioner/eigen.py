@@ -92,6 +92,11 @@ class EigenComputeMixin:
                 for i, r in zip(big, tracked_eig_multi(
                         trackers, [mats[i] for i in big])):
                     results[i] = r
+                if hasattr(self, "phase_times"):  # KFAC_PHASE_TIMING
+                    self.phase_times["eig_warm_total"] = float(
+                        sum(t.warm_count for t in trackers))
+                    self.phase_times["eig_cold_total"] = float(
+                        sum(t.cold_count for t in trackers))
         else:
             results = mat_eig_multi(mats, need_sorted=False)
 
