@@ -144,7 +144,15 @@ def tracked_eig_multi(trackers: List["EigenTracker"], mats) -> list:
         stats.append(trackers[i].prepare(mats[i]))
     oks = {}
     if stats:
+        import os
         host = torch.stack(stats).cpu()
+        if os.environ.get("KFAC_TRACKER_DEBUG"):
+            print("tracker stats (m, rel, s_rms, ok):",
+                  [(int(mats[i].shape[-1]), round(float(host[k, 0]), 4),
+                    round(float(host[k, 1]), 4),
+                    bool(float(host[k, 0]) <= trackers[i].cold_tol
+                         and float(host[k, 1]) <= trackers[i].s_tol))
+                   for k, i in enumerate(warm)], flush=True)
         for k, i in enumerate(warm):
             rel, s_rms = float(host[k, 0]), float(host[k, 1])
             oks[i] = (rel <= trackers[i].cold_tol
