@@ -167,7 +167,9 @@ def main():
 
     if rank == 0 and precond is not None and \
             hasattr(precond, "phase_times"):
-        per_step = {k: round(v / (args.steps + args.warmup) * 1000.0, 2)
+        # *_total entries are cumulative counts, the rest are seconds
+        per_step = {k: (round(v, 0) if k.endswith("_total") else
+                        round(v / (args.steps + args.warmup) * 1000.0, 2))
                     for k, v in precond.phase_times.items()}
         print("KFAC_PHASES(ms/step):", json.dumps(per_step),
               file=sys.stderr, flush=True)
