@@ -43,17 +43,23 @@ __all__ = ["EigenTracker", "tracked_eig_multi"]
 
 
 class EigenTracker:
-    def __init__(self, cold_every: int = 50, cold_tol: float = 3e-2,
+    def __init__(self, cold_every: int = 50, cold_tol: float = 0.1,
                  s_tol: float = 0.08, gap_rel: float = 1e-3,
-                 gap_abs: float = 1e-12):
+                 gap_floor_rel: float = 1e-6, gap_abs: float = 1e-12):
         # cold_tol: sqrt(gated off-mass / total mass) of B the correction
         #   is asked to remove; beyond it first-order is invalid.
-        # s_tol: RMS of S; beyond it (I+S) strays too far from
-        #   orthogonal for one Newton-Schulz step.
+        # s_tol: RMS of (clamped) S; beyond it (I+S) strays too far from
+        #   orthogonal for two Newton-Schulz steps.
+        # gap_floor_rel: ABSOLUTE gap floor as a fraction of the largest
+        #   eigenvalue -- gaps below ~1e-6*d_max are beneath fp32
+        #   eigensolver resolution, so K-FAC spectra's near-zero bulk
+        #   (rank-deficient sample covariances) is treated as one
+        #   cluster instead of generating unbounded rotations.
         self.cold_every = cold_every
         self.cold_tol = cold_tol
         self.s_tol = s_tol
         self.gap_rel = gap_rel
+        self.gap_floor_rel = gap_floor_rel
         self.gap_abs = gap_abs
         self.Q: Optional[torch.Tensor] = None
         self.calls_since_cold = 0
@@ -87,12 +93,17 @@ class EigenTracker:
         d = B.diagonal()
         gap = d.unsqueeze(0) - d.unsqueeze(1)      # gap[i,j] = d_j - d_i
         thr = (self.gap_rel * (d.abs().unsqueeze(0) + d.abs().unsqueeze(1))
+               + self.gap_floor_rel * d.abs().max()
                + self.gap_abs)
         mask = gap.abs() > thr
         S = torch.where(mask,
                         B / torch.where(mask, gap, torch.ones_like(gap)),
                         torch.zeros_like(B))
         S.fill_diagonal_(0.0)
+        # damped-Jacobi style clamp: oversized first-order rotations are
+        # applied partially (still reduces off-mass; keeps I+S in the
+        # Newton-Schulz convergence basin)
+        S.clamp_(-0.25, 0.25)
         m = A.shape[-1]
         off2 = (B * B * mask.to(B.dtype)).sum()
         tot2 = (B * B).sum().clamp_min(1e-30)
