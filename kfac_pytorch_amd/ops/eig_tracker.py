@@ -44,17 +44,22 @@ __all__ = ["EigenTracker", "tracked_eig_multi"]
 
 class EigenTracker:
     def __init__(self, cold_every: int = 50, cold_tol: float = 0.1,
-                 s_tol: float = 0.08, gap_rel: float = 1e-3,
-                 gap_floor_rel: float = 1e-6, gap_abs: float = 1e-12):
+                 s_tol: float = 0.15, gap_rel: float = 0.03,
+                 gap_floor_rel: float = 1e-5, gap_abs: float = 1e-12):
         # cold_tol: sqrt(gated off-mass / total mass) of B the correction
         #   is asked to remove; beyond it first-order is invalid.
         # s_tol: RMS of (clamped) S; beyond it (I+S) strays too far from
         #   orthogonal for two Newton-Schulz steps.
-        # gap_floor_rel: ABSOLUTE gap floor as a fraction of the largest
-        #   eigenvalue -- gaps below ~1e-6*d_max are beneath fp32
-        #   eigensolver resolution, so K-FAC spectra's near-zero bulk
-        #   (rank-deficient sample covariances) is treated as one
-        #   cluster instead of generating unbounded rotations.
+        # gap_rel: mixing tolerance.  Rotating eigenvectors whose
+        #   eigenvalues differ by Delta changes the K-FAC denominator
+        #   1/(dG*dA + damping) by at most Delta/(d + damping/dG_max)
+        #   relative -- a RELATIVE gap criterion; leaving pairs with
+        #   <= 3% eigenvalue difference mixed bounds the preconditioner
+        #   error at ~3%, far below the statistical noise of the factor
+        #   estimate itself.
+        # gap_floor_rel: absolute floor as a fraction of d_max -- the
+        #   near-zero bulk of rank-deficient sample covariances is one
+        #   cluster (also beneath fp32 eigensolver resolution).
         self.cold_every = cold_every
         self.cold_tol = cold_tol
         self.s_tol = s_tol
