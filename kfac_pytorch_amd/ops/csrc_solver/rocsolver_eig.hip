@@ -62,12 +62,19 @@ struct Slot {
   rocblas_handle handle = nullptr;
   hipStream_t stream = nullptr;
   hipEvent_t event = nullptr;
+  void* workspace = nullptr;
 };
 
 constexpr int POOL = 8;
+// Persistent per-handle device workspace.  Without it rocBLAS
+// auto-allocates (hipMalloc/hipFree) inside every rocSOLVER call --
+// device-synchronizing operations that serialize the whole pool and
+// kill cross-stream overlap.  256 MB covers ssyevd up to m ~ 8k.
+constexpr size_t WORKSPACE_BYTES = size_t(256) << 20;
 
 Slot g_pool[POOL];
 rocblas_handle g_main_handle = nullptr;  // bound to torch current stream
+void* g_main_workspace = nullptr;
 hipEvent_t g_acq_event = nullptr;
 bool g_init = false;
 
@@ -80,8 +87,15 @@ void ensure_init() {
     HIPCHECK(hipEventCreateWithFlags(&g_pool[i].event,
                                      hipEventDisableTiming));
     ROCBLASCHECK(rocblas_set_stream(g_pool[i].handle, g_pool[i].stream));
+    HIPCHECK(hipMalloc(&g_pool[i].workspace, WORKSPACE_BYTES));
+    ROCBLASCHECK(rocblas_set_workspace(g_pool[i].handle,
+                                       g_pool[i].workspace,
+                                       WORKSPACE_BYTES));
   }
   ROCBLASCHECK(rocblas_create_handle(&g_main_handle));
+  HIPCHECK(hipMalloc(&g_main_workspace, WORKSPACE_BYTES));
+  ROCBLASCHECK(rocblas_set_workspace(g_main_handle, g_main_workspace,
+                                     WORKSPACE_BYTES));
   HIPCHECK(hipEventCreateWithFlags(&g_acq_event, hipEventDisableTiming));
   g_init = true;
 }
