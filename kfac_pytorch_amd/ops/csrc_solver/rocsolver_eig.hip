@@ -105,8 +105,11 @@ void ensure_init() {
 // Batched eigensolve of same-size matrices: As (b, n, n) fp32 contiguous,
 // OVERWRITTEN with eigenvectors (column-major -> row i = eigenvector i
 // when read row-major).  Returns W (b, n) ascending and the info tensor
-// (b,) for deferred validation.  Runs on the torch current stream.
-std::vector<torch::Tensor> syevdj_batched_(torch::Tensor As) {
+// (b,) for deferred validation.  ``slot`` >= 0 issues the call on that
+// pool stream (event-ordered after the torch stream; caller must
+// ``join_pool_`` before consuming) so multiple dim-groups and the
+// mixed-size singles overlap; slot < 0 runs on the torch current stream.
+std::vector<torch::Tensor> syevdj_batched_(torch::Tensor As, int slot) {
   ensure_init();
   TORCH_CHECK(As.is_cuda() && As.dim() == 3 && As.size(1) == As.size(2),
               "syevdj_batched_: (b, n, n) GPU tensor required");
@@ -117,13 +120,38 @@ std::vector<torch::Tensor> syevdj_batched_(torch::Tensor As) {
   auto W = at::empty({b, n}, As.options());
   auto info = at::empty({b}, As.options().dtype(at::kInt));
   auto stream = c10::hip::getCurrentHIPStream();
-  ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  if (slot < 0) {
+    ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+    ROCBLASCHECK(rocsolver_ssyevdj_strided_batched(
+        g_main_handle, rocblas_evect_original, rocblas_fill_lower,
+        (rocblas_int)n, As.data_ptr<float>(), (rocblas_int)n,
+        (rocblas_stride)(n * n), W.data_ptr<float>(), (rocblas_stride)n,
+        info.data_ptr<int>(), (rocblas_int)b));
+    return {W, info};
+  }
+  Slot& s = g_pool[slot % POOL];
+  HIPCHECK(hipEventRecord(g_acq_event, stream.stream()));
+  HIPCHECK(hipStreamWaitEvent(s.stream, g_acq_event, 0));
+  record_on(As, s.stream);
+  record_on(W, s.stream);
+  record_on(info, s.stream);
   ROCBLASCHECK(rocsolver_ssyevdj_strided_batched(
-      g_main_handle, rocblas_evect_original, rocblas_fill_lower,
+      s.handle, rocblas_evect_original, rocblas_fill_lower,
       (rocblas_int)n, As.data_ptr<float>(), (rocblas_int)n,
       (rocblas_stride)(n * n), W.data_ptr<float>(), (rocblas_stride)n,
       info.data_ptr<int>(), (rocblas_int)b));
   return {W, info};
+}
+
+// Make the torch current stream wait on every pool stream (call after a
+// burst of slot-issued syevdj/syevd work).
+void join_pool_() {
+  ensure_init();
+  auto stream = c10::hip::getCurrentHIPStream();
+  for (int i = 0; i < POOL; ++i) {
+    HIPCHECK(hipEventRecord(g_pool[i].event, g_pool[i].stream));
+    HIPCHECK(hipStreamWaitEvent(stream.stream(), g_pool[i].event, 0));
+  }
 }
 
 // Pool-parallel eigensolves of differently-sized matrices: each matrix
@@ -168,8 +196,9 @@ std::vector<torch::Tensor> syevd_pool_(std::vector<torch::Tensor> mats) {
         info.data_ptr<int>() + i));
     out.push_back(W);
   }
-  // join: torch stream waits on all pool streams
-  for (int i = 0; i < POOL && i < (int)mats.size(); ++i) {
+  // join: torch stream waits on all pool streams (all of them -- the
+  // caller may have issued group work on slots this call didn't use)
+  for (int i = 0; i < POOL; ++i) {
     HIPCHECK(hipEventRecord(g_pool[i].event, g_pool[i].stream));
     HIPCHECK(hipStreamWaitEvent(stream.stream(), g_pool[i].event, 0));
   }
@@ -220,7 +249,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "async/batched rocSOLVER eigensolves for K-FAC (MI355X)";
   m.def("syevdj_batched_", &syevdj_batched_,
         "in-place batched symmetric eigensolve of (b,n,n); returns "
-        "(W, info); eigenvectors left row-major-transposed in input");
+        "(W, info); eigenvectors left row-major-transposed in input; "
+        "slot >= 0 issues on that pool stream (join_pool_ after)",
+        py::arg("As"), py::arg("slot") = -1);
+  m.def("join_pool_", &join_pool_,
+        "torch current stream waits on all pool streams");
   m.def("syevd_pool_", &syevd_pool_,
         "in-place pool-stream-overlapped eigensolves of mixed sizes; "
         "returns [W..., info]");

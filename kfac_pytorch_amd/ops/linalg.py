@@ -130,12 +130,18 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
         for i in rest:
             groups.setdefault(int(mats[i].shape[-1]), []).append(i)
         singles = []
-        for n, idxs in sorted(groups.items()):
+        slot = 0
+        issued_on_pool = False
+        for n, idxs in sorted(groups.items(), reverse=True):
             if len(idxs) < 2 or n > SYEVDJ_MAX:
                 singles.extend(idxs)
                 continue
             stacked = torch.stack([mats[i] for i in idxs]).contiguous()
-            W, _info = solver.syevdj_batched_(stacked)
+            # issue each dim-group on its own pool stream so groups
+            # overlap each other and the mixed-size syevd singles below
+            W, _info = solver.syevdj_batched_(stacked, slot)
+            slot += 1
+            issued_on_pool = True
             for k, i in enumerate(idxs):
                 # rocSOLVER leaves eigenvectors column-major in the
                 # buffer: row-major row i = eigenvector i, so .mT gives
@@ -146,9 +152,11 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
             singles.sort(key=lambda i: -mats[i].shape[-1])
             work = [mats[i].clone(
                 memory_format=torch.contiguous_format) for i in singles]
-            res = solver.syevd_pool_(work)
+            res = solver.syevd_pool_(work)  # joins ALL pool streams
             for k, i in enumerate(singles):
                 out[i] = (res[k], work[k].mT)
+        elif issued_on_pool:
+            solver.join_pool_()
         return out
 
     for i in rest:
