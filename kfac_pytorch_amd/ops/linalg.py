@@ -139,7 +139,13 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
             stacked = torch.stack([mats[i] for i in idxs]).contiguous()
             # issue each dim-group on its own pool stream so groups
             # overlap each other and the mixed-size syevd singles below
-            W, _info = solver.syevdj_batched_(stacked, slot)
+            try:
+                W, _info = solver.syevdj_batched_(stacked, slot)
+            except RuntimeError:
+                # workspace overflow for a very large group -> reroute
+                # the group members through the syevd pool
+                singles.extend(idxs)
+                continue
             slot += 1
             issued_on_pool = True
             for k, i in enumerate(idxs):
