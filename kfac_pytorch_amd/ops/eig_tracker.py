@@ -117,15 +117,14 @@ class EigenTracker:
         self._pending = (B, d, S)
         return stats
 
-    def commit(self, A: torch.Tensor,
-               ok: bool) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Apply the prepared warm update if ``ok``, else cold-restart."""
+    def apply_(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Apply the prepared (clamped) rotation to the tracked basis;
+        returns the corrected (eigenvalues, Q).  Does NOT touch the
+        warm/cold counters -- callers iterate this until the health
+        stats go green (quadratic convergence near the solution, the
+        clamp makes distant starts contract too)."""
         B, d, S = self._pending
         self._pending = None
-        if not ok:
-            return self._cold(A)
-        self.calls_since_cold += 1
-        self.warm_count += 1
         Q = self.Q
         Qn = Q + Q @ S
         for _ in range(2):  # Newton-Schulz: error cubes per step
@@ -135,54 +134,80 @@ class EigenTracker:
         d_corr = d - (B * S).sum(dim=1)
         return d_corr, Qn
 
-    def update(self, A: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Single-factor convenience wrapper (one host sync)."""
+    def commit(self, A: torch.Tensor,
+               ok: bool) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Finalize one tracked update: apply if ``ok``, else cold."""
+        if not ok:
+            self._pending = None
+            return self._cold(A)
+        self.calls_since_cold += 1
+        self.warm_count += 1
+        return self.apply_()
+
+    def is_green(self, stats) -> bool:
+        rel, s_rms = float(stats[0]), float(stats[1])
+        return rel <= self.cold_tol and s_rms <= self.s_tol
+
+    def update(self, A: torch.Tensor, max_rounds: int = 4
+               ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Single-factor convenience wrapper (one host sync per round)."""
         if self.needs_cold(A):
             return self._cold(A)
-        stats = self.prepare(A)
-        rel, s_rms = (float(x) for x in stats.cpu())
-        return self.commit(A, ok=(rel <= self.cold_tol
-                                  and s_rms <= self.s_tol))
+        result = None
+        for _ in range(max_rounds):
+            stats = self.prepare(A).cpu()
+            green = self.is_green(stats)
+            result = self.apply_()
+            if green:
+                self.calls_since_cold += 1
+                self.warm_count += 1
+                return result
+        return self._cold(A)
 
 
-def tracked_eig_multi(trackers: List["EigenTracker"], mats) -> list:
-    """Eigendecompose many factors with warm tracking: all warm
-    candidates' GPU work is queued first, ONE host transfer reads every
-    health stat, then each factor commits warm or joins the batched cold
-    solve.  Returns [(w, Q), ...] aligned with ``mats``."""
+def tracked_eig_multi(trackers: List["EigenTracker"], mats,
+                      max_rounds: int = 4) -> list:
+    """Eigendecompose many factors with warm ITERATIVE tracking: each
+    round queues every active factor's correction on-stream, ONE host
+    transfer reads all health stats, green factors finish, red ones
+    iterate (refined basis) up to ``max_rounds``; whatever is left joins
+    the batched cold solve.  Returns [(w, Q), ...] aligned with
+    ``mats``."""
+    import os
     from kfac_pytorch_amd.ops.linalg import mat_eig_multi
+    debug = os.environ.get("KFAC_TRACKER_DEBUG")
     out = [None] * len(mats)
     cold = [i for i, t in enumerate(trackers) if t.needs_cold(mats[i])]
-    warm = [i for i in range(len(mats)) if i not in set(cold)]
+    active = [i for i in range(len(mats)) if i not in set(cold)]
 
-    stats = []
-    for i in warm:
-        stats.append(trackers[i].prepare(mats[i]))
-    oks = {}
-    if stats:
-        import os
+    for rnd in range(max_rounds):
+        if not active:
+            break
+        stats = [trackers[i].prepare(mats[i]) for i in active]
         host = torch.stack(stats).cpu()
-        if os.environ.get("KFAC_TRACKER_DEBUG"):
-            print("tracker stats (m, rel, s_rms, ok):",
+        if debug:
+            print(f"tracker round {rnd} (m, rel, s_rms, green):",
                   [(int(mats[i].shape[-1]), round(float(host[k, 0]), 4),
                     round(float(host[k, 1]), 4),
-                    bool(float(host[k, 0]) <= trackers[i].cold_tol
-                         and float(host[k, 1]) <= trackers[i].s_tol))
-                   for k, i in enumerate(warm)], flush=True)
-        for k, i in enumerate(warm):
-            rel, s_rms = float(host[k, 0]), float(host[k, 1])
-            oks[i] = (rel <= trackers[i].cold_tol
-                      and s_rms <= trackers[i].s_tol)
-            if not oks[i]:
-                cold.append(i)
+                    trackers[i].is_green(host[k]))
+                   for k, i in enumerate(active)], flush=True)
+        still = []
+        for k, i in enumerate(active):
+            green = trackers[i].is_green(host[k])
+            result = trackers[i].apply_()
+            if green:
+                trackers[i].calls_since_cold += 1
+                trackers[i].warm_count += 1
+                out[i] = result
+            else:
+                still.append(i)
+        active = still
 
+    cold.extend(active)  # unconverged after max_rounds
     if cold:
         solved = mat_eig_multi([mats[i] for i in cold], need_sorted=False)
         for i, (w, Q) in zip(cold, solved):
             trackers[i]._pending = None
             trackers[i].seed(w, Q.clone())
             out[i] = (w, Q)
-    for i in warm:
-        if out[i] is None:
-            out[i] = trackers[i].commit(mats[i], ok=True)
     return out

@@ -48,9 +48,9 @@ class EigenComputeMixin:
     ``KFAC_EIG_TRACKER=0`` disables it.
     """
 
-    #: factors above this dim use the perturbative tracker (the
-    #: syevdj-batched tier below it is already ~ms-cheap)
-    TRACK_MIN = 768
+    #: factors above this dim use the iterative perturbative tracker
+    #: (below it the batched LDS-Jacobi / syevdj tiers are sub-ms)
+    TRACK_MIN = 192
 
     def _eigendecompose_owned(self):
         import os
