@@ -193,11 +193,20 @@ def tracked_eig_multi(trackers: List["EigenTracker"], mats,
                    for k, i in enumerate(active)], flush=True)
         still = []
         for k, i in enumerate(active):
-            green = trackers[i].is_green(host[k])
-            result = trackers[i].apply_()
+            tr = trackers[i]
+            green = tr.is_green(host[k])
+            hopeless = (float(host[k, 0]) > 4 * tr.cold_tol
+                        or float(host[k, 1]) > 4 * tr.s_tol)
+            if hopeless and not green:
+                # basis moved too far for iterative refinement to pay
+                # off -- skip the wasted rounds, go straight to cold
+                tr._pending = None
+                cold.append(i)
+                continue
+            result = tr.apply_()
             if green:
-                trackers[i].calls_since_cold += 1
-                trackers[i].warm_count += 1
+                tr.calls_since_cold += 1
+                tr.warm_count += 1
                 out[i] = result
             else:
                 still.append(i)
