@@ -315,3 +315,39 @@ def _worker_factor_wise(rank, world, tmpfile):
 
 def test_factor_wise_distribution():
     _run_spawn(_worker_factor_wise)
+
+
+# --------------------------------------------------------------------------
+def _worker_inverse_bcast_mode(rank, world, tmpfile):
+    """KFACInverse with communicate_inverse_or_not=True: inverses are
+    broadcast and every rank preconditions locally; results must equal
+    the default pred-broadcast mode exactly
+    (reference: kfac_preconditioner_inv.py:41,132-142)."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    torch.manual_seed(21)
+    m1 = MLP()
+    m2 = MLP()
+    m2.load_state_dict(m1.state_dict())
+    for p in list(m1.parameters()) + list(m2.parameters()):
+        comm.broadcast(p.data, src=0)
+    p1 = kfac.KFAC_INV(m1, damping=0.01, communicate_inverse_or_not=True)
+    p2 = kfac.KFAC_INV(m2, damping=0.01, communicate_inverse_or_not=False)
+    x, y = _global_batch(seed=77)
+    half = x.shape[0] // world
+    xs = x[rank * half:(rank + 1) * half]
+    ys = y[rank * half:(rank + 1) * half]
+    for step in range(2):
+        for mod, pre in ((m1, p1), (m2, p2)):
+            _train_grads(mod, xs, ys)
+            for p in mod.parameters():
+                comm.allreduce(p.grad.data, op=comm.Average)
+            pre.step()
+        for a, b in zip(m1.parameters(), m2.parameters()):
+            torch.testing.assert_close(a.grad, b.grad,
+                                       rtol=1e-5, atol=1e-6)
+    dist.destroy_process_group()
+
+
+def test_inverse_broadcast_mode_matches_pred_mode():
+    _run_spawn(_worker_inverse_bcast_mode)
