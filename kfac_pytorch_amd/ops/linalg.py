@@ -133,14 +133,20 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
         slot = 0
         issued_on_pool = False
         for n, idxs in sorted(groups.items(), reverse=True):
-            if len(idxs) < 2 or n > SYEVDJ_MAX:
+            if len(idxs) < 2:
                 singles.extend(idxs)
                 continue
             stacked = torch.stack([mats[i] for i in idxs]).contiguous()
             # issue each dim-group on its own pool stream so groups
-            # overlap each other and the mixed-size syevd singles below
+            # overlap each other and the mixed-size syevd singles below.
+            # Jacobi (syevdj) wins below SYEVDJ_MAX; above it batched
+            # divide-and-conquer fills the latency gaps of the
+            # column-by-column tridiagonalization panels.
             try:
-                W, _info = solver.syevdj_batched_(stacked, slot)
+                if n > SYEVDJ_MAX:
+                    W, _info = solver.syevd_batched_(stacked, slot)
+                else:
+                    W, _info = solver.syevdj_batched_(stacked, slot)
             except RuntimeError:
                 # workspace overflow for a very large group -> reroute
                 # the group members through the syevd pool

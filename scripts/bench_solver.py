@@ -59,7 +59,8 @@ def main():
     print(f"magma available = {has_magma}")
 
     hdr = (f"{'m':>6} {'eigh':>9} {'magma':>9} {'pool/8':>9} "
-           f"{'syevdj/8':>10} {'jac/8':>9} {'potri/8':>9} {'chol_inv':>9}")
+           f"{'syevdj/8':>10} {'syevdB/8':>10} {'jac/8':>9} "
+           f"{'potri/8':>9} {'chol_inv':>9}")
     print(hdr + "   (ms per matrix)")
     for m in DIMS:
         a = spd(m)
@@ -83,7 +84,18 @@ def main():
         def sjb():
             w = batch.clone()
             solver.syevdj_batched_(w)
-        t_sjb = timeit(sjb) / b
+        try:
+            t_sjb = timeit(sjb) / b
+        except RuntimeError:
+            t_sjb = float("nan")
+
+        def sdb():
+            w = batch.clone()
+            solver.syevd_batched_(w)
+        try:
+            t_sdb = timeit(sdb) / b
+        except RuntimeError:
+            t_sdb = float("nan")
 
         t_jac = float("nan")
         if _ext.has_jacobi_eigh(m):
@@ -100,7 +112,8 @@ def main():
             lambda: torch.cholesky_inverse(torch.linalg.cholesky(a)))
 
         print(f"{m:>6} {t_eigh:9.2f} {t_magma:9.2f} {t_pool:9.2f} "
-              f"{t_sjb:10.2f} {t_jac:9.2f} {t_potri:9.2f} {t_chol:9.2f}")
+              f"{t_sjb:10.2f} {t_sdb:10.2f} {t_jac:9.2f} "
+              f"{t_potri:9.2f} {t_chol:9.2f}")
 
 
 if __name__ == "__main__":
