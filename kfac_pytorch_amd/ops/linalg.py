@@ -74,9 +74,34 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
 
 # dispatch cutoffs, measured on MI355X (profiles/bench_solver.log):
 # hand-written LDS-Jacobi wins at m <= 64 (0.33 ms/matrix batched);
-# rocSOLVER syevdj_strided_batched wins 65..1152; async-pool syevd above.
+# batched divide-and-conquer syevd wins everywhere above (3-10x the
+# single-matrix path: its tiny tridiagonalization panel kernels only
+# fill the chip when batched).  PAD_RATIO buckets nearby dims into one
+# padded batch (pad block = -1 diagonal, sorted out exactly below the
+# PSD spectrum): <= 15% dim padding (~50% flops on the smallest
+# member) buys another 3x+ of batching.
 JAC_DISPATCH_MAX = 64
 SYEVDJ_MAX = 1152
+PAD_RATIO = 1.16
+
+
+def _pad_buckets(dims):
+    """Greedy bucketing of sorted-desc (dim, idx) pairs: each bucket's
+    members are within PAD_RATIO of the leader and get padded to it."""
+    buckets = []
+    cur = []
+    lead = None
+    for d, i in dims:
+        if lead is None or d * PAD_RATIO >= lead:
+            if lead is None:
+                lead = d
+            cur.append((d, i))
+        else:
+            buckets.append((lead, cur))
+            lead, cur = d, [(d, i)]
+    if cur:
+        buckets.append((lead, cur))
+    return buckets
 
 
 def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
@@ -122,43 +147,42 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
             and _solver_ok():
         from kfac_pytorch_amd.ops import _ext
         solver = _ext.load_solver()
-        # group same-dim matrices -> one batched syevdj call per group
-        # (measured on MI355X: syevdj_strided_batched wins up to
-        # m~1152, collapses beyond -- 1.45 s at m=4608; big matrices go
-        # to the async syevd pool, profiles/bench_solver)
-        groups = {}
-        for i in rest:
-            groups.setdefault(int(mats[i].shape[-1]), []).append(i)
+        dims = sorted(((int(mats[i].shape[-1]), i) for i in rest),
+                      reverse=True)
         singles = []
         slot = 0
         issued_on_pool = False
-        for n, idxs in sorted(groups.items(), reverse=True):
-            if len(idxs) < 2:
-                singles.extend(idxs)
+        device = mats[rest[0]].device
+        for n, members in _pad_buckets(dims):
+            if len(members) < 2:
+                singles.extend(i for _, i in members)
                 continue
-            stacked = torch.stack([mats[i] for i in idxs]).contiguous()
-            # issue each dim-group on its own pool stream so groups
-            # overlap each other and the mixed-size syevd singles below.
-            # Jacobi (syevdj) wins below SYEVDJ_MAX; above it batched
-            # divide-and-conquer fills the latency gaps of the
-            # column-by-column tridiagonalization panels.
+            b = len(members)
+            stacked = torch.full((b, n, n), 0.0, device=device)
+            for k, (m, i) in enumerate(members):
+                pad = n - m
+                stacked[k, pad:, pad:] = mats[i]
+                if pad:
+                    # isolated pad block: -1 diagonal sorts strictly
+                    # below the PSD factor spectrum, so the pad
+                    # eigenpairs are exactly W[:pad] after the
+                    # ascending-order solve
+                    stacked[k].diagonal()[:pad] = -1.0
             try:
-                if n > SYEVDJ_MAX:
-                    W, _info = solver.syevd_batched_(stacked, slot)
-                else:
-                    W, _info = solver.syevdj_batched_(stacked, slot)
+                W, _info = solver.syevd_batched_(stacked, slot)
             except RuntimeError:
-                # workspace overflow for a very large group -> reroute
-                # the group members through the syevd pool
-                singles.extend(idxs)
+                singles.extend(i for _, i in members)
                 continue
             slot += 1
             issued_on_pool = True
-            for k, i in enumerate(idxs):
+            for k, (m, i) in enumerate(members):
+                pad = n - m
                 # rocSOLVER leaves eigenvectors column-major in the
                 # buffer: row-major row i = eigenvector i, so .mT gives
-                # the eigh contract (columns = eigenvectors)
-                out[i] = (W[k], stacked[k].mT)
+                # the eigh contract (columns = eigenvectors); the real
+                # eigenpairs are the top m columns / bottom-right rows
+                Q = stacked[k].mT
+                out[i] = (W[k, pad:], Q[pad:, pad:] if pad else Q)
         if singles:
             # largest-first keeps the 8 pool streams balanced
             singles.sort(key=lambda i: -mats[i].shape[-1])

@@ -181,3 +181,20 @@ def test_tracker_rank_deficient_noisy_stays_accurate(solver):
         w, Q = tracker.update(A)
         err = _precond_err(A, w, Q)
         assert err < 0.12, (t, err)
+
+
+def test_mat_eig_multi_padded_buckets(solver):
+    """Nearby dims are padded into one batched syevd (pad block = -1
+    diagonal): results must still match eigh exactly on the real part."""
+    from kfac_pytorch_amd.ops.linalg import mat_eig_multi
+    dims = [2048, 2049, 2304, 1000, 1024, 1152, 513, 512, 576]
+    mats = [spd(m, seed=m) for m in dims]
+    out = mat_eig_multi(mats, need_sorted=False)
+    torch.cuda.synchronize()
+    for a, (w, Q) in zip(mats, out):
+        assert w.shape[0] == a.shape[0]
+        assert Q.shape == a.shape
+        check_eig(a, w, Q)
+        w_ref = torch.linalg.eigvalsh(a)
+        torch.testing.assert_close(torch.sort(w).values, w_ref,
+                                   rtol=2e-3, atol=2e-3)
