@@ -69,11 +69,11 @@ constexpr int POOL = 8;
 // Persistent per-handle device workspace.  Without it rocBLAS
 // auto-allocates (hipMalloc/hipFree) inside every rocSOLVER call --
 // device-synchronizing operations that serialize the whole pool and
-// kill cross-stream overlap.  1 GB per handle (9 GB total of the
-// 288 GB HBM) covers ssyevd at m ~ 8k and batched syevdj groups; a
+// kill cross-stream overlap.  4 GB per handle (36 GB total of the
+// 288 GB HBM) covers batched ssyevd groups at m ~ 4.6k; a
 // request beyond it surfaces as rocblas_status_memory_error, which the
 // Python dispatch catches and reroutes to the syevd pool.
-constexpr size_t WORKSPACE_BYTES = size_t(1) << 30;
+constexpr size_t WORKSPACE_BYTES = size_t(4) << 30;
 
 Slot g_pool[POOL];
 rocblas_handle g_main_handle = nullptr;  // bound to torch current stream

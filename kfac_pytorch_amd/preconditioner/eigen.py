@@ -38,14 +38,12 @@ class EigenComputeMixin:
     Replaces the reference's serial per-layer eigh loop
     (reference: kfac_preconditioner_eigen.py:98-119).
 
-    Factors larger than ``TRACK_MIN`` additionally go through the
-    health-gated perturbative eigenbasis tracker (ops/eig_tracker.py):
-    when the factor drifts slowly between updates the full library
-    solve (118 ms at m=4608) is replaced by ~7 GEMMs; the per-factor
-    health gate falls back to the exact solve whenever first-order
-    tracking is invalid (e.g. heavy batch noise), so results stay
-    within the gate tolerance of the exact decomposition always.
-    ``KFAC_EIG_TRACKER=0`` disables it.
+    ``KFAC_EIG_TRACKER=1`` additionally routes factors larger than
+    ``TRACK_MIN`` through the health-gated perturbative eigenbasis
+    tracker (ops/eig_tracker.py) -- worthwhile in steady-state training
+    where factors drift slowly; off by default because the batched
+    cold solves above are already within ~2x of the tracker's warm
+    path and exact (measured: profiles/PERFORMANCE.md).
     """
 
     #: factors above this dim use the iterative perturbative tracker
@@ -67,7 +65,7 @@ class EigenComputeMixin:
         mats = [self.m_A[mod] if kind == "A" else self.m_G[mod]
                 for mod, kind in work]
 
-        use_tracker = (os.environ.get("KFAC_EIG_TRACKER", "1") != "0"
+        use_tracker = (os.environ.get("KFAC_EIG_TRACKER", "0") == "1"
                        and mats[0].is_cuda)
         if use_tracker:
             from kfac_pytorch_amd.ops.eig_tracker import (EigenTracker,
