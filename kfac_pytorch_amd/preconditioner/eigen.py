@@ -25,14 +25,18 @@ from kfac_pytorch_amd.preconditioner.inverse import KFACInverse
 
 
 class EigenComputeMixin:
-    """Shared owner-side eigendecomposition, MI355X-scheduled:
+    """Shared owner-side eigendecomposition, MI355X-scheduled
+    (dispatch measured in profiles/bench_solver.log):
 
-    * small factors (<= the LDS-Jacobi cap) batch into ONE hand-written
+    * small factors (m <= 64) batch into ONE hand-written LDS-Jacobi
       kernel launch;
-    * same-dim factor groups batch into one rocSOLVER
-      ``syevdj_strided_batched`` call; leftover singles overlap on a
-      persistent async stream pool (no host syncs -- torch.linalg.eigh
-      would host-sync per matrix and serialize the whole set);
+    * everything else buckets by dim (within ~15%, padded with an
+      isolated -1 diagonal block) into strided-batched rocSOLVER
+      divide-and-conquer eigensolves -- 3-10x the single-matrix path,
+      whose tiny tridiagonalization panels only fill the chip when
+      batched; leftover singletons overlap on a persistent async
+      stream pool (no host syncs -- torch.linalg.eigh would host-sync
+      per matrix and serialize the whole set);
     * CPU falls back to plain eigh.
 
     Replaces the reference's serial per-layer eigh loop
