@@ -17,6 +17,8 @@ from __future__ import annotations
 
 from typing import Tuple
 
+import os
+
 import torch
 
 __all__ = [
@@ -193,6 +195,25 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                 out[i] = (res[k], work[k].mT)
         elif issued_on_pool:
             solver.join_pool_()
+        if os.environ.get("KFAC_EIG_DEBUG"):
+            torch.cuda.synchronize()
+            import time as _t
+            rep = []
+            for n, members in _pad_buckets(dims):
+                t0 = _t.perf_counter()
+                if len(members) < 2:
+                    w = [mats[i].clone() for _, i in members]
+                    solver.syevd_pool_(w)
+                else:
+                    b = len(members)
+                    st = torch.full((b, n, n), 0.0, device=device)
+                    for k, (m, i) in enumerate(members):
+                        st[k, n - m:, n - m:] = mats[i]
+                    solver.syevd_batched_(st, -1)
+                torch.cuda.synchronize()
+                rep.append((n, len(members),
+                            round((_t.perf_counter() - t0) * 1e3, 1)))
+            print("eig buckets (n, count, ms):", rep, flush=True)
         return out
 
     for i in rest:
