@@ -110,6 +110,25 @@ class KFACBase(optim.Optimizer):
     def comm(self):
         return comm_mod.get_comm()
 
+    def _native_comm(self):
+        """Opt-in (KFAC_NATIVE_COMM=1) multi-stream RCCL communicator
+        for the bucket collectives: duplicate RCCL comms on their own
+        HIP streams so broadcasts rooted at different owners ride
+        different xGMI links (ops/csrc_rccl).  Lazily and collectively
+        created at first use; None when disabled / CPU / world 1."""
+        if not hasattr(self, "_native_comm_obj"):
+            self._native_comm_obj = None
+            import os
+            if (os.environ.get("KFAC_NATIVE_COMM") == "1"
+                    and torch.cuda.is_available()
+                    and self.comm.size() > 1):
+                from kfac_pytorch_amd.parallel.native import (
+                    NativeCommunicator, native_available)
+                if native_available():
+                    self._native_comm_obj = NativeCommunicator.create(
+                        num_comms=min(4, self.comm.size()))
+        return self._native_comm_obj
+
     # ----------------------------------------------------------------- hooks
     def set_hook_enabled(self, mode: bool = True):
         self.hook_enabled = mode
@@ -247,15 +266,28 @@ class KFACBase(optim.Optimizer):
         """One allreduce-average for a whole phase's tensors."""
         if bucket.buffer is None or bucket.buffer.numel() == 0:
             return
+        nat = self._native_comm()
+        if nat is not None:
+            nat.all_reduce(bucket.buffer, average=True)
+            nat.join()
+            return
         self.comm.allreduce(bucket.buffer, op=self.comm.Average)
 
     def _broadcast_owner_buckets(self, buckets: List[FlatBucket]):
         """Async broadcast of each owner's flat bucket from its rank,
-        issued concurrently on rotating duplicate process groups so
+        issued concurrently on rotating duplicate communicators so
         different roots ride different xGMI links; then drain.
 
         Replaces the reference's per-layer broadcast bursts
         (kfac/kfac_preconditioner_inv.py:132-142,164-175)."""
+        nat = self._native_comm()
+        if nat is not None:
+            for r, b in enumerate(buckets):
+                if b.buffer is None or len(b) == 0:
+                    continue
+                nat.broadcast(b.buffer, root=r)
+            nat.join()  # stream-ordered, no host block
+            return
         c = self.comm
         c.ensure_rotating_groups()
         handles = []
