@@ -51,11 +51,13 @@ class EigenComputeMixin:
     """
 
     #: factors above this dim use the iterative perturbative tracker
-    #: (below it the batched LDS-Jacobi / syevdj tiers are sub-ms)
+    #: when it is enabled (below it the batched solver tiers are cheap);
+    #: override with KFAC_TRACK_MIN
     TRACK_MIN = 192
 
     def _eigendecompose_owned(self):
         import os
+        track_min = int(os.environ.get("KFAC_TRACK_MIN", self.TRACK_MIN))
         rank = self.comm.rank()
         work = []
         for m in self.modules:
@@ -77,7 +79,7 @@ class EigenComputeMixin:
             if not hasattr(self, "_eig_trackers"):
                 self._eig_trackers = {}
             small = [i for i, a in enumerate(mats)
-                     if a.shape[-1] <= self.TRACK_MIN]
+                     if a.shape[-1] <= track_min]
             big = [i for i in range(len(mats)) if i not in set(small)]
             results = [None] * len(mats)
             if small:
