@@ -286,6 +286,42 @@ std::vector<torch::Tensor> potri_pool_(std::vector<torch::Tensor> mats) {
   return {info};
 }
 
+// Batched Cholesky inverse of same-size SPD matrices (potrf + potri,
+// strided-batched): As (b, n, n) fp32 contiguous, OVERWRITTEN with the
+// inverse's column-major LOWER triangle = row-major UPPER triangle
+// (mirror in the caller).  slot semantics as syevd_batched_.
+std::vector<torch::Tensor> potri_batched_(torch::Tensor As, int slot) {
+  ensure_init();
+  TORCH_CHECK(As.is_cuda() && As.dim() == 3 && As.size(1) == As.size(2),
+              "potri_batched_: (b, n, n) GPU tensor required");
+  TORCH_CHECK(As.scalar_type() == at::kFloat && As.is_contiguous(),
+              "potri_batched_: fp32 contiguous required");
+  const long b = As.size(0);
+  const long n = As.size(1);
+  auto info = at::empty({2 * b}, As.options().dtype(at::kInt));
+  auto stream = c10::hip::getCurrentHIPStream();
+  rocblas_handle h = g_main_handle;
+  if (slot >= 0) {
+    Slot& s = g_pool[slot % POOL];
+    HIPCHECK(hipEventRecord(g_acq_event, stream.stream()));
+    HIPCHECK(hipStreamWaitEvent(s.stream, g_acq_event, 0));
+    record_on(As, s.stream);
+    record_on(info, s.stream);
+    h = s.handle;
+  } else {
+    ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  }
+  ROCBLASCHECK(rocsolver_spotrf_strided_batched(
+      h, rocblas_fill_lower, (rocblas_int)n, As.data_ptr<float>(),
+      (rocblas_int)n, (rocblas_stride)(n * n), info.data_ptr<int>(),
+      (rocblas_int)b));
+  ROCBLASCHECK(rocsolver_spotri_strided_batched(
+      h, rocblas_fill_lower, (rocblas_int)n, As.data_ptr<float>(),
+      (rocblas_int)n, (rocblas_stride)(n * n),
+      info.data_ptr<int>() + b, (rocblas_int)b));
+  return {info};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "async/batched rocSOLVER eigensolves for K-FAC (MI355X)";
   m.def("syevdj_batched_", &syevdj_batched_,
@@ -304,4 +340,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("potri_pool_", &potri_pool_,
         "in-place pool-stream-overlapped Cholesky inverse (lower) of "
         "mixed sizes; returns [info]");
+  m.def("potri_batched_", &potri_batched_,
+        "in-place strided-batched Cholesky inverse of (b,n,n); "
+        "returns [info]", py::arg("As"), py::arg("slot") = -1);
 }

@@ -248,11 +248,48 @@ def mat_inv_multi(mats, damp_diag=None):
     if mats[0].is_cuda and _solver_ok():
         from kfac_pytorch_amd.ops import _ext
         solver = _ext.load_solver()
-        solver.potri_pool_(work)
-        # potri fills only the column-major lower triangle = our
-        # row-major UPPER triangle (the row-major lower still holds the
-        # damped input); mirror the upper triangle down
-        return [w.triu(0) + w.triu(1).mT for w in work]
+        out = [None] * len(work)
+        dims = sorted(((int(w.shape[-1]), i) for i, w in enumerate(work)),
+                      reverse=True)
+        singles = []
+        slot = 0
+        issued = False
+        pending = []
+        for n, members in _pad_buckets(dims):
+            if len(members) < 2:
+                singles.extend(i for _, i in members)
+                continue
+            b = len(members)
+            stacked = torch.zeros(b, n, n, device=work[0].device)
+            for k, (m, i) in enumerate(members):
+                pad = n - m
+                stacked[k, pad:, pad:] = work[i]
+                if pad:
+                    stacked[k].diagonal()[:pad] = 1.0  # SPD identity pad
+            try:
+                solver.potri_batched_(stacked, slot)
+            except RuntimeError:
+                singles.extend(i for _, i in members)
+                continue
+            slot += 1
+            issued = True
+            pending.append((stacked, members))
+        if singles:
+            sw = [work[i] for i in singles]
+            solver.potri_pool_(sw)  # joins ALL pool streams
+        elif issued:
+            solver.join_pool_()
+        # mirror AFTER the join so the torch stream sees finished potri
+        for stacked, members in pending:
+            n = stacked.shape[-1]
+            for k, (m, i) in enumerate(members):
+                pad = n - m
+                w = stacked[k, pad:, pad:]
+                out[i] = w.triu(0) + w.triu(1).mT
+        for i in singles:
+            w = work[i]
+            out[i] = w.triu(0) + w.triu(1).mT
+        return out
     return [mat_inv(w) for w in work]
 
 
