@@ -14,13 +14,13 @@ issued concurrently on rotating RCCL communicators.
 
 from __future__ import annotations
 
-from typing import Dict, List
+from typing import Dict
 
 import torch
 import torch.nn as nn
 
 from kfac_pytorch_amd.ops.factors import ComputeA, ComputeG, factor_dims
-from kfac_pytorch_amd.ops.linalg import (add_diagonal_, inverse_precondition,
+from kfac_pytorch_amd.ops.linalg import (inverse_precondition,
                                          mat_inv)
 from kfac_pytorch_amd.preconditioner.base import KFACBase
 

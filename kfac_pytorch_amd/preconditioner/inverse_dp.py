@@ -16,8 +16,7 @@ from __future__ import annotations
 
 import torch
 
-from kfac_pytorch_amd.ops.linalg import (add_diagonal_, inverse_precondition,
-                                         mat_inv)
+from kfac_pytorch_amd.ops.linalg import inverse_precondition
 from kfac_pytorch_amd.preconditioner.inverse import KFACInverse
 
 
