@@ -130,10 +130,14 @@ __device__ __forceinline__ void stage_tile(
 __global__ __launch_bounds__(256) void syrk_bf16_kernel(
     const unsigned short* __restrict__ X, float* __restrict__ Ctmp,
     long rows, int d, long ldx, int n, int bias, int nsplit, long ksplit,
-    int ntiles) {
-  // blockIdx.x -> (upper-triangle tile pair, K split)
+    int ntiles, int nslabs) {
+  // blockIdx.x -> (upper-triangle tile pair, K split); splits spread
+  // their atomics over nslabs separate accumulation slabs so hundreds
+  // of split-K blocks do not hammer one tiny n x n output (d <= 128
+  // factors showed ~4x atomic-contention slowdowns)
   const int tp = blockIdx.x / nsplit;
   const int split = blockIdx.x - tp * nsplit;
+  Ctmp += (long)(split % nslabs) * n * n;
   // tile pair tp -> (ti, tj) with ti <= tj, row-major over the triangle
   int ti = 0, rem = tp;
   while (rem >= ntiles - ti) {
@@ -218,10 +222,15 @@ __global__ __launch_bounds__(256) void syrk_bf16_kernel(
 
 __global__ void syrk_epilogue_kernel(float* __restrict__ out,
                                      const float* __restrict__ tmp,
-                                     float alpha, float decay, long total) {
+                                     float alpha, float decay, long total,
+                                     int nslabs) {
   for (long idx = (long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long)gridDim.x * blockDim.x) {
-    const float fresh = tmp[idx] * alpha;
+    float acc = 0.f;
+    for (int s = 0; s < nslabs; ++s) {
+      acc += tmp[s * total + idx];
+    }
+    const float fresh = acc * alpha;
     out[idx] = (decay < 0.f) ? fresh
                              : (1.f - decay) * out[idx] + decay * fresh;
   }
@@ -344,7 +353,6 @@ torch::Tensor syrk_factor_entry(torch::Tensor x, torch::Tensor out,
   TORCH_CHECK(rows > 0, "syrk_factor_: empty input");
 
   auto stream = c10::hip::getCurrentHIPStream();
-  auto tmp = at::zeros_like(out);
 
   const int ntiles = (n + BT - 1) / BT;
   const long npairs = (long)ntiles * (ntiles + 1) / 2;
@@ -355,12 +363,16 @@ torch::Tensor syrk_factor_entry(torch::Tensor x, torch::Tensor out,
   if (nsplit < 1) nsplit = 1;
   long ksplit = (rows + nsplit - 1) / nsplit;
   ksplit = (ksplit + BK - 1) / BK * BK;
+  // spread split-K atomics over up to 16 accumulation slabs (bounded
+  // extra memory: 16 * n^2 floats, n <= a few K)
+  const int nslabs = (int)std::min<long>(nsplit, 16);
 
+  auto tmp = at::zeros({(long)nslabs, (long)n, (long)n}, out.options());
   const long grid = npairs * nsplit;
   syrk_bf16_kernel<<<(int)grid, 256, 0, stream.stream()>>>(
       reinterpret_cast<const unsigned short*>(x.data_ptr<at::BFloat16>()),
       tmp.data_ptr<float>(), rows, d, x.stride(0), n, bias ? 1 : 0,
-      (int)nsplit, ksplit, ntiles);
+      (int)nsplit, ksplit, ntiles, nslabs);
   HIP_CHECK(hipGetLastError());
 
   const float alpha = (float)(row_scale * row_scale / denom);
@@ -368,7 +380,7 @@ torch::Tensor syrk_factor_entry(torch::Tensor x, torch::Tensor out,
   const int grid2 = (int)std::min<long>((total + 255) / 256, 2048);
   syrk_epilogue_kernel<<<grid2, 256, 0, stream.stream()>>>(
       out.data_ptr<float>(), tmp.data_ptr<float>(), alpha, (float)decay,
-      total);
+      total, nslabs);
   HIP_CHECK(hipGetLastError());
   return out;
 }
