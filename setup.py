@@ -31,7 +31,8 @@ setup(
     packages=find_packages(include=["kfac_pytorch_amd",
                                     "kfac_pytorch_amd.*"]),
     package_data={"kfac_pytorch_amd.ops": ["*.so", "csrc/*.hip",
-                                           "csrc_rccl/*.hip"]},
+                                           "csrc_rccl/*.hip",
+                                           "csrc_solver/*.hip"]},
     cmdclass={"build_ext": HipBuildExt},
     python_requires=">=3.9",
 )

@@ -66,3 +66,20 @@ def test_inverse_precondition():
     out = inverse_precondition(torch.linalg.inv(A), torch.linalg.inv(G), grad)
     expected = torch.linalg.inv(G) @ grad @ torch.linalg.inv(A)
     torch.testing.assert_close(out, expected, rtol=1e-4, atol=1e-5)
+
+
+def test_pad_buckets_grouping():
+    """Bucketed-padding grouping: members within PAD_RATIO of the
+    leader share a bucket (ResNet-50's 2048/2049/2304 and
+    1000/1024/1152 must merge)."""
+    from kfac_pytorch_amd.ops.linalg import _pad_buckets
+    dims = sorted([(4608, 0), (2304, 1), (2049, 2), (2048, 3),
+                   (1152, 4), (1024, 5), (1000, 6), (576, 7), (512, 8)],
+                  reverse=True)
+    buckets = _pad_buckets(dims)
+    as_sets = [(lead, sorted(i for _, i in members))
+               for lead, members in buckets]
+    assert as_sets[0] == (4608, [0])
+    assert as_sets[1] == (2304, [1, 2, 3])
+    assert as_sets[2] == (1152, [4, 5, 6])
+    assert as_sets[3] == (576, [7, 8])
