@@ -162,6 +162,16 @@ class KFACBase(optim.Optimizer):
                 continue  # exclude pre-softmax vocab projection (ref :139-140)
             if classname == 'Conv2d' and module.groups != 1:
                 continue  # grouped conv factors don't match weight layout
+            if (classname == 'Linear' and module.out_features >= 10000
+                    and self.exclude_vocabulary_size is None):
+                logger.warning(
+                    "KFAC: hooking Linear with out_features=%d -- its G "
+                    "factor alone is %d MB fp32; pass "
+                    "exclude_vocabulary_size=%d to skip vocab-sized "
+                    "projections (reference behavior for LM heads)",
+                    module.out_features,
+                    module.out_features ** 2 * 4 // 2 ** 20,
+                    module.out_features)
             self.modules.append(module)
             self._hook_handles.append(
                 module.register_forward_pre_hook(self._forward_hook_event))

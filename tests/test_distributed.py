@@ -351,3 +351,44 @@ def _worker_inverse_bcast_mode(rank, world, tmpfile):
 
 def test_inverse_broadcast_mode_matches_pred_mode():
     _run_spawn(_worker_inverse_bcast_mode)
+
+
+# --------------------------------------------------------------------------
+def _worker_world4_eigen(rank, world, tmpfile):
+    """World-4 MPD eigen: round-robin owners across 4 ranks, eigenbases
+    broadcast on rotating groups, identical results on every rank --
+    the shape of the driver's 8-GPU run at CPU scale."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    torch.manual_seed(31)
+    model = MLP()
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+    pre = kfac.KFAC_EIGEN(model, damping=0.01)
+    x, y = _global_batch(seed=5, n=16)
+    quarter = x.shape[0] // world
+    xs = x[rank * quarter:(rank + 1) * quarter]
+    ys = y[rank * quarter:(rank + 1) * quarter]
+    for step in range(2):
+        _train_grads(model, xs, ys)
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        pre.step()
+        # world(4) > modules(3) -> factor-wise distribution: A and G of
+        # each layer on different ranks (reference: eigen.py:66-71)
+        assert all(ra != rg for ra, rg in pre.module_ranks.values())
+        used = {r for pair in pre.module_ranks.values() for r in pair}
+        assert len(used) >= 3
+        for p in model.parameters():
+            mine = p.grad.clone()
+            comm.broadcast(p.grad.data, src=0)
+            torch.testing.assert_close(mine, p.grad, rtol=1e-5,
+                                       atol=1e-6)
+    dist.destroy_process_group()
+
+
+def test_world4_eigen_consistency():
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(_worker_world4_eigen, args=(4, tmpfile), nprocs=4, join=True)
