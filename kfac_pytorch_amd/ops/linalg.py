@@ -6,26 +6,39 @@ plus the two preconditioning chains
 (reference: kfac/kfac_preconditioner_inv.py:156-161 /
 kfac_preconditioner_eigen.py:137-144).
 
-GPU dispatch: the symmetric eigensolver has a hand-written CDNA4 HIP
-path (batched Jacobi) for factor-sized matrices; larger matrices use
-``torch.linalg.eigh`` (rocSOLVER) behind the same ``method`` switch so
-correctness never blocks on the kernel.  The eigen-precondition scale
-``V / (dG dA^T + damping)`` is a fused elementwise HIP kernel on GPU.
+GPU dispatch (all cutoffs measured, profiles/bench_solver.log):
+
+* m <= 64: hand-written batched LDS-Jacobi HIP kernel, one launch for
+  a whole rank's small factors;
+* everything else: bucketed by dim within ~15% (padded with an
+  isolated diagonal block) into strided-batched rocSOLVER
+  divide-and-conquer eigensolves / Cholesky inversions driven through
+  ``_kfac_solver`` (persistent handles + 4 GB device workspaces, pool
+  streams, zero host syncs) -- 3-10x the per-matrix library path;
+* singletons overlap on the 8-stream async pool;
+* CPU falls back to ``torch.linalg.eigh`` / Cholesky so correctness
+  never blocks on a kernel (same ``method`` switch as the reference,
+  kfac/utils.py:22-30).
+
+The eigen-precondition scale ``V / (dG dA^T + damping)`` is a fused
+elementwise HIP kernel on GPU (batched across same-shape layers).
 """
 
 from __future__ import annotations
 
-from typing import Tuple
-
 import os
+from typing import Tuple
 
 import torch
 
 __all__ = [
     "add_diagonal_",
     "mat_inv",
+    "mat_inv_multi",
     "mat_eig",
+    "mat_eig_multi",
     "eigen_precondition",
+    "eigen_precondition_multi",
     "inverse_precondition",
 ]
 
