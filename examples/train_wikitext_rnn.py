@@ -42,8 +42,14 @@ def parse_args():
     p.add_argument("--layers", type=int, default=2)
     p.add_argument("--bptt", type=int, default=35,
                    help="sequence length (reference default)")
-    p.add_argument("--exclude-vocab", action="store_true",
-                   help="exclude the vocab-sized decoder from K-FAC")
+    p.add_argument("--exclude-vocab", dest="exclude_vocab",
+                   action="store_true", default=True,
+                   help="exclude the vocab-sized decoder from K-FAC "
+                        "(default: a 33k-dim factor is not worth "
+                        "eigendecomposing; reference LM trainers exclude "
+                        "vocab projections too)")
+    p.add_argument("--include-vocab", dest="exclude_vocab",
+                   action="store_false")
     p.set_defaults(batch_size=20, base_lr=20.0 / 20, damping=0.003)
     return p.parse_args()
 

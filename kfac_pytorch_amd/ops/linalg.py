@@ -203,9 +203,17 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
             singles.sort(key=lambda i: -mats[i].shape[-1])
             work = [mats[i].clone(
                 memory_format=torch.contiguous_format) for i in singles]
-            res = solver.syevd_pool_(work)  # joins ALL pool streams
-            for k, i in enumerate(singles):
-                out[i] = (res[k], work[k].mT)
+            try:
+                res = solver.syevd_pool_(work)  # joins ALL pool streams
+                for k, i in enumerate(singles):
+                    out[i] = (res[k], work[k].mT)
+            except RuntimeError:
+                # a matrix larger than the persistent workspace covers
+                # (e.g. an un-excluded vocab-sized factor): fall back to
+                # torch's own eigh for the whole singles set
+                solver.join_pool_()
+                for i in singles:
+                    out[i] = mat_eig(mats[i], method="eigh")
         elif issued_on_pool:
             solver.join_pool_()
         if os.environ.get("KFAC_EIG_DEBUG"):

@@ -369,7 +369,7 @@ class KFACBase(optim.Optimizer):
                 p.grad.data.copy_(vi)
                 grads.append(p.grad.data)
 
-        if use_clip:
+        if use_clip and vg_sum is not None:
             if self.exclude_communicate_inverse:
                 return  # nu == 1 (reference :209-212)
             nu = torch.clamp(

@@ -276,3 +276,21 @@ def test_kfac_converges_on_toy_problem(single_process_comm, seeded, name):
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < 0.5 * losses[0], (losses[0], losses[-1])
+
+
+def test_zero_hooked_modules_step_is_noop(single_process_comm, seeded):
+    """A model whose only preconditionable layer is vocab-excluded (the
+    WikiText LSTM default) must step without error and leave grads
+    untouched."""
+    from kfac_pytorch_amd.models import LSTMLanguageModel
+    m = LSTMLanguageModel(vocab_size=211, emb=16, hidden=16, layers=1)
+    pre = kfac.get_kfac_module("eigen_dp")(m, exclude_vocabulary_size=211)
+    assert len(pre.modules) == 0
+    out, _ = m(torch.randint(0, 211, (2, 4)))
+    out.sum().backward()
+    before = [p.grad.clone() for p in m.parameters()
+              if p.grad is not None]
+    pre.step()
+    after = [p.grad for p in m.parameters() if p.grad is not None]
+    for b, a in zip(before, after):
+        torch.testing.assert_close(b, a)
