@@ -322,6 +322,60 @@ std::vector<torch::Tensor> potri_batched_(torch::Tensor As, int slot) {
   return {info};
 }
 
+// The three stages of a symmetric eigensolve, exposed separately so
+// the Python side can (a) profile the sytrd/stedc/ormtr split and
+// (b) swap any stage for a custom kernel (round-2 SBR plan: replace
+// sytrd, keep library D&C + back-transform).  All run on the torch
+// current stream via the main handle.
+std::vector<torch::Tensor> sytrd_(torch::Tensor A) {
+  ensure_init();
+  TORCH_CHECK(A.is_cuda() && A.dim() == 2 && A.size(0) == A.size(1) &&
+                  A.scalar_type() == at::kFloat && A.is_contiguous(),
+              "sytrd_: square fp32 contiguous GPU matrix required");
+  const long n = A.size(0);
+  auto D = at::empty({n}, A.options());
+  auto E = at::empty({n}, A.options());
+  auto tau = at::empty({n}, A.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  ROCBLASCHECK(rocsolver_ssytrd(g_main_handle, rocblas_fill_lower,
+                                (rocblas_int)n, A.data_ptr<float>(),
+                                (rocblas_int)n, D.data_ptr<float>(),
+                                E.data_ptr<float>(),
+                                tau.data_ptr<float>()));
+  return {D, E, tau};
+}
+
+std::vector<torch::Tensor> stedc_(torch::Tensor D, torch::Tensor E) {
+  ensure_init();
+  const long n = D.numel();
+  TORCH_CHECK(D.is_cuda() && E.is_cuda() && E.numel() >= n - 1,
+              "stedc_: device D/E required");
+  auto C = at::eye(n, D.options());
+  auto info = at::empty({1}, D.options().dtype(at::kInt));
+  auto stream = c10::hip::getCurrentHIPStream();
+  ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  ROCBLASCHECK(rocsolver_sstedc(g_main_handle, rocblas_evect_original,
+                                (rocblas_int)n, D.data_ptr<float>(),
+                                E.data_ptr<float>(), C.data_ptr<float>(),
+                                (rocblas_int)n, info.data_ptr<int>()));
+  return {C, info};
+}
+
+void ormtr_(torch::Tensor A, torch::Tensor tau, torch::Tensor C) {
+  ensure_init();
+  const long n = A.size(0);
+  TORCH_CHECK(C.size(0) == n && C.size(1) == n && C.is_contiguous(),
+              "ormtr_: C must be (n, n) contiguous");
+  auto stream = c10::hip::getCurrentHIPStream();
+  ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  ROCBLASCHECK(rocsolver_sormtr(
+      g_main_handle, rocblas_side_left, rocblas_fill_lower,
+      rocblas_operation_none, (rocblas_int)n, (rocblas_int)n,
+      A.data_ptr<float>(), (rocblas_int)n, tau.data_ptr<float>(),
+      C.data_ptr<float>(), (rocblas_int)n));
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "async/batched rocSOLVER eigensolves for K-FAC (MI355X)";
   m.def("syevdj_batched_", &syevdj_batched_,
@@ -334,6 +388,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "contract as syevdj_batched_", py::arg("As"), py::arg("slot") = -1);
   m.def("join_pool_", &join_pool_,
         "torch current stream waits on all pool streams");
+  m.def("sytrd_", &sytrd_, "in-place tridiagonalization; returns D,E,tau");
+  m.def("stedc_", &stedc_,
+        "tridiagonal D&C eigensolve; returns (C eigvec col-major, info)");
+  m.def("ormtr_", &ormtr_,
+        "C <- Q C with Q from sytrd reflectors (in-place)");
   m.def("syevd_pool_", &syevd_pool_,
         "in-place pool-stream-overlapped eigensolves of mixed sizes; "
         "returns [W..., info]");
