@@ -83,3 +83,40 @@ def test_pad_buckets_grouping():
     assert as_sets[1] == (2304, [1, 2, 3])
     assert as_sets[2] == (1152, [4, 5, 6])
     assert as_sets[3] == (576, [7, 8])
+
+
+def test_eigen_precondition_equals_dense_kronecker_solve():
+    """The implicit-eigen chain QG^T g QA / (dG dA^T + l) -> QG V QA^T
+    must equal solving the dense damped Kronecker system
+    (G (x) A + l I) vec(P) = vec(grad) directly -- the core K-FAC
+    identity the whole eigen family rests on."""
+    torch.manual_seed(4)
+    da, dg, lam = 7, 5, 0.013
+    xa = torch.randn(da, da, dtype=torch.float64)
+    xg = torch.randn(dg, dg, dtype=torch.float64)
+    A = xa @ xa.T / da + 0.1 * torch.eye(da, dtype=torch.float64)
+    G = xg @ xg.T / dg + 0.1 * torch.eye(dg, dtype=torch.float64)
+    grad = torch.randn(dg, da, dtype=torch.float64)
+
+    dA, QA = torch.linalg.eigh(A)
+    dG, QG = torch.linalg.eigh(G)
+    P = eigen_precondition(QA, dA, QG, dG, grad.clone(), lam)
+
+    # dense oracle: row-major vec with K = kron(G, A) matching
+    # vec(G P A) = (G kron A^T)... use the transpose identity:
+    # G P A = reshape((G kron A^T) vec_r(P)); A symmetric -> A^T = A
+    K = torch.kron(G, A) + lam * torch.eye(dg * da, dtype=torch.float64)
+    P_ref = torch.linalg.solve(K, grad.reshape(-1)).reshape(dg, da)
+
+    torch.testing.assert_close(P, P_ref, rtol=1e-8, atol=1e-10)
+
+
+def test_eigen_clamp_matches_reference_formula():
+    """Eigenvalue clamp d*(d>eps) (reference
+    kfac_preconditioner_base.py:115 / eigen.py:114,119) zeroes tiny and
+    negative eigenvalues before the denominator."""
+    d = torch.tensor([-1e-3, 0.0, 1e-12, 1e-9, 0.5])
+    eps = 1e-10
+    clamped = d * (d > eps)
+    assert torch.equal(clamped,
+                       torch.tensor([0.0, 0.0, 0.0, 1e-9, 0.5]))
