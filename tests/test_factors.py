@@ -166,3 +166,35 @@ def test_conv_factor_matches_naive_conv_gradient(seeded):
     w = layer.weight.view(layer.out_channels, -1)
     y2 = (patches @ w.t()).view(2, y.size(2), y.size(3), 6).permute(0, 3, 1, 2)
     torch.testing.assert_close(y, y2, rtol=1e-4, atol=1e-5)
+
+
+def test_extract_patches_asymmetric_kernel():
+    """(1,7)/(7,1) factorized convs (inception family) produce correct
+    patch rows on the CPU oracle path."""
+    torch.manual_seed(8)
+    x = torch.randn(2, 3, 9, 11)
+    for ks, pad in [((1, 7), (0, 3)), ((7, 1), (3, 0)), ((3, 5), (1, 2))]:
+        got = extract_patches(x, ks, (1, 1), pad)
+        cols = F.unfold(x, kernel_size=ks, stride=(1, 1), padding=pad)
+        ref = cols.transpose(1, 2).reshape(-1, cols.size(1))
+        torch.testing.assert_close(got, ref)
+
+
+def test_conv_factor_asymmetric_kernel_shapes(single_process_comm):
+    """K-FAC steps through a model with (1,7)/(7,1) convs -- factor
+    dims include kh*kw correctly."""
+    import kfac_pytorch_amd as kfac
+    m = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 4, (1, 7), padding=(0, 3)),
+        torch.nn.ReLU(),
+        torch.nn.Conv2d(4, 5, (7, 1), padding=(3, 0)),
+        torch.nn.AdaptiveAvgPool2d(1), torch.nn.Flatten(),
+        torch.nn.Linear(5, 3))
+    pre = kfac.KFAC_EIGEN_DP(m, damping=0.01)
+    from kfac_pytorch_amd.ops.factors import factor_dims
+    da0, dg0 = factor_dims(m[0])
+    assert (da0, dg0) == (3 * 7 + 1, 4)
+    out = m(torch.randn(2, 3, 9, 9))
+    F.cross_entropy(out, torch.tensor([0, 1])).backward()
+    pre.step()
+    assert all(torch.isfinite(p.grad).all() for p in m.parameters())
