@@ -53,3 +53,23 @@ def test_train_transformer_small():
         ["--batch-size", "4", "--iters-per-epoch", "2", "--epochs", "1",
          "--speed", "--display", "1"])
     assert "images/s" in out
+
+
+def test_train_imagenet_checkpoint_resume(tmp_path):
+    """Save a checkpoint at epoch 0, resume from it: the resume path
+    (utils.load_checkpoint + KFACParamScheduler epoch) must work
+    (reference: examples/pytorch_imagenet_resnet.py:162-167,305-312)."""
+    ckpt = str(tmp_path / "ck-{epoch}.pth.tar")
+    common_args = ["--model", "resnet18", "--num-classes", "7",
+                   "--image-size", "64", "--batch-size", "2",
+                   "--iters-per-epoch", "2", "--display", "1",
+                   "--kfac-update-freq", "2"]
+    run_example("examples/train_imagenet.py",
+                common_args + ["--epochs", "1",
+                               "--checkpoint-format", ckpt])
+    saved = str(tmp_path / "ck-0.pth.tar")
+    assert os.path.exists(saved)
+    out = run_example("examples/train_imagenet.py",
+                      common_args + ["--epochs", "2",
+                                     "--resume-from", saved])
+    assert "epoch 1" in out  # resumed into epoch 1
