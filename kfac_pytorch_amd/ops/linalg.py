@@ -96,7 +96,6 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
 # PSD spectrum): <= 15% dim padding (~50% flops on the smallest
 # member) buys another 3x+ of batching.
 JAC_DISPATCH_MAX = 64
-SYEVDJ_MAX = 1152
 PAD_RATIO = 1.16
 
 
@@ -122,13 +121,13 @@ def _pad_buckets(dims):
 def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
     """Eigendecompose a list of symmetric matrices the MI355X way:
 
-    * every Jacobi-eligible matrix (m <= 128) goes into ONE hand-written
-      LDS-Jacobi kernel launch;
-    * remaining same-dim groups batch into ONE rocSOLVER
-      ``syevdj_strided_batched`` call (K-FAC factor dims repeat heavily
-      across a network);
-    * leftover singles are issued async on a persistent handle+stream
-      pool (``syevd_pool_``) and joined once.
+    * every Jacobi-eligible matrix (m <= JAC_DISPATCH_MAX) goes into
+      ONE hand-written LDS-Jacobi kernel launch;
+    * the rest is bucketed by dim within PAD_RATIO (padded with an
+      isolated -1 diagonal block) into strided-batched rocSOLVER
+      divide-and-conquer eigensolves, one bucket per pool slot;
+    * leftover singletons are issued async on a persistent
+      handle+stream pool (``syevd_pool_``) and joined once.
 
     torch.linalg.eigh host-syncs per call, so the per-layer loop the
     reference runs serializes ~106 eigensolves per ResNet-50 step
