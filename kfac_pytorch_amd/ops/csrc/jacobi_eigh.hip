@@ -205,45 +205,6 @@ __global__ __launch_bounds__(256) void jacobi_eigh_kernel(
 
 int jacobi_eigh_max_dim() { return JAC_MAX_DIM; }
 
-// Uniform-batch entry: B[nb, b, b] -> (W[nb, b], V[nb, b, b]) in one
-// launch with no per-matrix packing (the hot path of the block-Jacobi
-// eigen tracker: all diagonal 64-blocks / paired 128-blocks of every
-// tracked factor eigensolve together).
-std::vector<torch::Tensor> jacobi_eigh_batched_uniform(torch::Tensor B) {
-  TORCH_CHECK(B.is_cuda() && B.dim() == 3 && B.size(1) == B.size(2),
-              "jacobi_eigh_batched_uniform: (nb, b, b) GPU tensor only");
-  TORCH_CHECK(B.scalar_type() == at::kFloat && B.is_contiguous(),
-              "jacobi_eigh_batched_uniform: fp32 contiguous only");
-  const long nb = B.size(0);
-  const int b = (int)B.size(1);
-  TORCH_CHECK(b <= JAC_MAX_DIM, "jacobi_eigh_batched_uniform: dim ", b,
-              " > ", JAC_MAX_DIM);
-  auto opts = B.options();
-  auto W = at::empty({nb, (long)b}, opts);
-  auto V = at::empty_like(B);
-  std::vector<long> mat_off(nb), vec_off(nb);
-  std::vector<int> sizes(nb, b);
-  for (long i = 0; i < nb; ++i) {
-    mat_off[i] = i * (long)b * b;
-    vec_off[i] = i * (long)b;
-  }
-  auto lopts = at::TensorOptions().dtype(at::kLong);
-  auto iopts = at::TensorOptions().dtype(at::kInt);
-  const auto dev = B.device();
-  auto mat_off_d = at::from_blob(mat_off.data(), {nb}, lopts).to(dev);
-  auto vec_off_d = at::from_blob(vec_off.data(), {nb}, lopts).to(dev);
-  auto sizes_d = at::from_blob(sizes.data(), {nb}, iopts).to(dev);
-  auto stream = c10::hip::getCurrentHIPStream();
-  jacobi_eigh_kernel<<<(int)nb, 256, 0, stream.stream()>>>(
-      B.data_ptr<float>(), W.data_ptr<float>(), V.data_ptr<float>(),
-      mat_off_d.data_ptr<long>(), vec_off_d.data_ptr<long>(),
-      sizes_d.data_ptr<int>());
-  hipError_t err = hipGetLastError();
-  TORCH_CHECK(err == hipSuccess, "jacobi_eigh_batched_uniform launch: ",
-              hipGetErrorString(err));
-  return {W, V};
-}
-
 std::vector<torch::Tensor> jacobi_eigh_batched(
     std::vector<torch::Tensor> mats) {
   TORCH_CHECK(!mats.empty(), "jacobi_eigh_batched: empty batch");

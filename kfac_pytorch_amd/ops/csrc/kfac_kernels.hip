@@ -423,7 +423,6 @@ torch::Tensor im2col_entry(torch::Tensor x, long kh, long kw, long sh,
 int jacobi_eigh_max_dim();
 std::vector<torch::Tensor> jacobi_eigh_batched(
     std::vector<torch::Tensor> mats);
-std::vector<torch::Tensor> jacobi_eigh_batched_uniform(torch::Tensor B);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native HIP kernels for distributed K-FAC";
@@ -436,8 +435,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("im2col", &im2col_entry, "conv patch extraction to bf16 rows");
   m.def("jacobi_eigh_batched", &jacobi_eigh_batched,
         "batched LDS-resident Jacobi symmetric eigensolver (packed W, V)");
-  m.def("jacobi_eigh_batched_uniform", &jacobi_eigh_batched_uniform,
-        "uniform-batch Jacobi eigensolver: (nb,b,b) -> (W,V)");
   m.def("jacobi_eigh_max_dim", &jacobi_eigh_max_dim,
         "largest dim the Jacobi kernel handles");
 }

@@ -279,7 +279,11 @@ std::vector<torch::Tensor> potri_pool_(std::vector<torch::Tensor> mats) {
                                   (rocblas_int)n,
                                   info.data_ptr<int>() + 2 * i + 1));
   }
-  for (int i = 0; i < POOL && i < (int)mats.size(); ++i) {
+  // join ALL pool streams (not just the ones this call used): callers
+  // may have issued batched work on other slots before calling in, and
+  // they rely on "potri_pool_ joins the whole pool" before mirroring
+  // results on the torch stream.
+  for (int i = 0; i < POOL; ++i) {
     HIPCHECK(hipEventRecord(g_pool[i].event, g_pool[i].stream));
     HIPCHECK(hipStreamWaitEvent(stream.stream(), g_pool[i].event, 0));
   }

@@ -40,7 +40,42 @@ __all__ = [
     "eigen_precondition",
     "eigen_precondition_multi",
     "inverse_precondition",
+    "check_deferred_info",
 ]
+
+# ---------------------------------------------------------------- info flags
+# rocSOLVER writes per-matrix ``info`` words (non-convergence / non-SPD).
+# Checking them synchronously would host-sync the sync-free solve path,
+# so every batched/pool call accumulates info into one persistent device
+# counter per device; the preconditioner calls ``check_deferred_info()``
+# right before issuing the NEXT inverse/eigen update (the queue is
+# drained there anyway), raising at most one update late instead of
+# silently propagating garbage eigenpairs/inverses (the torch
+# eigh/cholesky path this replaces raised immediately).
+_INFO_FLAGS: dict = {}
+
+
+def _defer_info(info: torch.Tensor) -> None:
+    flag = _INFO_FLAGS.get(info.device)
+    if flag is None:
+        flag = torch.zeros((), dtype=torch.int64, device=info.device)
+        _INFO_FLAGS[info.device] = flag
+    flag.add_(info.ne(0).sum())
+
+
+def check_deferred_info() -> None:
+    """Raise if any rocSOLVER call since the last check reported failure
+    (Cholesky of a non-SPD factor / eigensolve non-convergence).
+    Host-syncs; call where the device queue is already drained."""
+    for dev, flag in _INFO_FLAGS.items():
+        bad = int(flag.item())
+        if bad:
+            flag.zero_()
+            raise RuntimeError(
+                f"{bad} rocSOLVER factorization(s) on {dev} failed since "
+                "the last check (non-SPD factor or eigensolve "
+                "non-convergence). A K-FAC update consumed those results; "
+                "raise damping or inspect the factors.")
 
 
 def add_diagonal_(X: torch.Tensor, value) -> torch.Tensor:
@@ -187,6 +222,7 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
             except RuntimeError:
                 singles.extend(i for _, i in members)
                 continue
+            _defer_info(_info)
             slot += 1
             issued_on_pool = True
             for k, (m, i) in enumerate(members):
@@ -204,6 +240,7 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                 memory_format=torch.contiguous_format) for i in singles]
             try:
                 res = solver.syevd_pool_(work)  # joins ALL pool streams
+                _defer_info(res[-1])
                 for k, i in enumerate(singles):
                     out[i] = (res[k], work[k].mT)
             except RuntimeError:
@@ -287,16 +324,28 @@ def mat_inv_multi(mats, damp_diag=None):
                 if pad:
                     stacked[k].diagonal()[:pad] = 1.0  # SPD identity pad
             try:
-                solver.potri_batched_(stacked, slot)
+                (_info,) = solver.potri_batched_(stacked, slot)
             except RuntimeError:
                 singles.extend(i for _, i in members)
                 continue
+            _defer_info(_info)
             slot += 1
             issued = True
             pending.append((stacked, members))
         if singles:
             sw = [work[i] for i in singles]
-            solver.potri_pool_(sw)  # joins ALL pool streams
+            try:
+                (_info,) = solver.potri_pool_(sw)  # joins ALL pool streams
+                _defer_info(_info)
+            except RuntimeError:
+                # a matrix beyond the persistent workspace (e.g. an
+                # un-excluded vocab-sized factor): degrade to the torch
+                # Cholesky path for the singles instead of aborting the
+                # whole inverse step (mirrors the eig-path fallback)
+                solver.join_pool_()
+                for i in singles:
+                    out[i] = mat_inv(work[i])
+                singles = []
         elif issued:
             solver.join_pool_()
         # mirror AFTER the join so the torch stream sees finished potri

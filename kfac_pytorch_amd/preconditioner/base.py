@@ -427,6 +427,11 @@ class KFACBase(optim.Optimizer):
 
         if self.steps % self.kfac_update_freq == 0:
             if not self.exclude_compute_inverse:
+                # surface any rocSOLVER failure from the PREVIOUS update
+                # before issuing new solves (deferred to keep the solve
+                # path sync-free; the queue is drained here anyway)
+                from kfac_pytorch_amd.ops.linalg import check_deferred_info
+                check_deferred_info()
                 with self._phase("compute_inverse"):
                     self._compute_inverse()
             if (not self.exclude_communicate_inverse
@@ -556,5 +561,11 @@ class KFACParamScheduler:
         params['damping'] = self.damping_base * \
             self.damping_factor_func(self.epoch)
         factor = self.update_freq_factor_func(self.epoch)
-        params['fac_update_freq'] = int(self.fac_update_freq_base * factor)
-        params['kfac_update_freq'] = int(self.kfac_update_freq_base * factor)
+        # clamp to >= 1: with update_freq_alpha < 1, int(base * factor)
+        # can reach 0 and crash the next ``steps % freq`` (the reference
+        # has this bug, kfac_preconditioner_base.py:288-301 -- fixed,
+        # not copied)
+        params['fac_update_freq'] = max(
+            1, int(self.fac_update_freq_base * factor))
+        params['kfac_update_freq'] = max(
+            1, int(self.kfac_update_freq_base * factor))

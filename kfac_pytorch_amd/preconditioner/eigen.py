@@ -41,23 +41,9 @@ class EigenComputeMixin:
 
     Replaces the reference's serial per-layer eigh loop
     (reference: kfac_preconditioner_eigen.py:98-119).
-
-    ``KFAC_EIG_TRACKER=1`` additionally routes factors larger than
-    ``TRACK_MIN`` through the health-gated perturbative eigenbasis
-    tracker (ops/eig_tracker.py) -- worthwhile in steady-state training
-    where factors drift slowly; off by default because the batched
-    cold solves above are already within ~2x of the tracker's warm
-    path and exact (measured: profiles/PERFORMANCE.md).
     """
 
-    #: factors above this dim use the iterative perturbative tracker
-    #: when it is enabled (below it the batched solver tiers are cheap);
-    #: override with KFAC_TRACK_MIN
-    TRACK_MIN = 192
-
     def _eigendecompose_owned(self):
-        import os
-        track_min = int(os.environ.get("KFAC_TRACK_MIN", self.TRACK_MIN))
         rank = self.comm.rank()
         work = []
         for m in self.modules:
@@ -70,39 +56,7 @@ class EigenComputeMixin:
             return
         mats = [self.m_A[mod] if kind == "A" else self.m_G[mod]
                 for mod, kind in work]
-
-        use_tracker = (os.environ.get("KFAC_EIG_TRACKER", "0") == "1"
-                       and mats[0].is_cuda)
-        if use_tracker:
-            from kfac_pytorch_amd.ops.eig_tracker import (EigenTracker,
-                                                          tracked_eig_multi)
-            if not hasattr(self, "_eig_trackers"):
-                self._eig_trackers = {}
-            small = [i for i, a in enumerate(mats)
-                     if a.shape[-1] <= track_min]
-            big = [i for i in range(len(mats)) if i not in set(small)]
-            results = [None] * len(mats)
-            if small:
-                for i, r in zip(small, mat_eig_multi(
-                        [mats[i] for i in small], need_sorted=False)):
-                    results[i] = r
-            if big:
-                # stagger periodic cold restarts across factors so no
-                # single step pays every library eigensolve at once
-                trackers = [self._eig_trackers.setdefault(
-                    (id(work[i][0]), work[i][1]),
-                    EigenTracker(cold_every=50 + 7 * (j % 13)))
-                    for j, i in enumerate(big)]
-                for i, r in zip(big, tracked_eig_multi(
-                        trackers, [mats[i] for i in big])):
-                    results[i] = r
-                if hasattr(self, "phase_times"):  # KFAC_PHASE_TIMING
-                    self.phase_times["eig_warm_total"] = float(
-                        sum(t.warm_count for t in trackers))
-                    self.phase_times["eig_cold_total"] = float(
-                        sum(t.cold_count for t in trackers))
-        else:
-            results = mat_eig_multi(mats, need_sorted=False)
+        results = mat_eig_multi(mats, need_sorted=False)
 
         for (mod, kind), (d, Q) in zip(work, results):
             clamped = d * (d > self.eps)

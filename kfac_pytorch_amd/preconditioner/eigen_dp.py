@@ -71,8 +71,18 @@ class KFACEigenDP(EigenComputeMixin, KFACInverseDP):
         (eigen_precondition_multi)."""
         assert not self.communicate_inverse_or_not
         rank = self.comm.rank()
-        owned = [m for m in self.modules
-                 if rank == self.module_ranks[m][0]]
+        owned = []
+        for m in self.modules:
+            rank_a, rank_g = self.module_ranks[m]
+            # the DP family requires one owner per layer; a factor-wise
+            # split (rank_a != rank_g) would silently skip the G-owner's
+            # preconditioning here -- fail loudly instead (reference
+            # asserts the same, kfac_preconditioner_eigen_dp.py:80)
+            assert rank_a == rank_g, (
+                "eigen_dp requires rank_a == rank_g per layer "
+                f"(got {rank_a} != {rank_g})")
+            if rank == rank_a:
+                owned.append(m)
         if not owned:
             return
         preds = eigen_precondition_multi(

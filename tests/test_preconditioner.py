@@ -177,6 +177,22 @@ def test_param_scheduler(single_process_comm, seeded):
     assert math.isclose(pre.damping, 0.0025)
 
 
+def test_param_scheduler_freq_never_zero(single_process_comm, seeded):
+    """update_freq_alpha < 1 must clamp the scheduled frequencies at 1:
+    int(base * factor) reaching 0 would crash ``steps % freq`` (bug in
+    the reference, kfac_preconditioner_base.py:288-301 -- fixed here)."""
+    model = TinyNet()
+    pre = kfac.KFAC_EIGEN_DP(model, fac_update_freq=2, kfac_update_freq=2)
+    sched = kfac.KFACParamScheduler(pre, update_freq_alpha=0.25,
+                                    update_freq_schedule=[1])
+    sched.step(epoch=1)
+    assert pre.param_groups[0]['fac_update_freq'] == 1
+    assert pre.param_groups[0]['kfac_update_freq'] == 1
+    run_fwd_bwd(model)
+    pre.step()  # must not divide by zero
+    assert pre.steps == 1
+
+
 def test_lambda_lr_compatibility(single_process_comm, seeded):
     """KFAC is an optim.Optimizer, so LambdaLR must drive its lr
     (reference usage: examples/pytorch_cifar10_resnet.py:276)."""
