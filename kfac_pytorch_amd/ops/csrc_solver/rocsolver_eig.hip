@@ -30,6 +30,7 @@
 #include <c10/hip/HIPStream.h>
 #include <c10/hip/HIPCachingAllocator.h>
 
+#include <algorithm>
 #include <sstream>
 #include <vector>
 
@@ -350,6 +351,123 @@ std::vector<torch::Tensor> sytrd_(torch::Tensor A) {
   return {D, E, tau};
 }
 
+// defined in sytrd_panel.hip
+std::vector<torch::Tensor> sytrd_panels_device(
+    torch::Tensor As, torch::Tensor W, torch::Tensor E,
+    torch::Tensor tau, torch::Tensor scratch, torch::Tensor syncc,
+    torch::Tensor status, int j0, int ib, int R, int wgs_alloc, int wgs);
+int sytrd_panel_max_blocks(int R);
+
+// Custom batched blocked tridiagonalization (sytrd_panel.hip): the
+// hand-written persistent-panel kernel replaces rocSOLVER's latrd
+// launch storm; the rank-2*ib trailing updates run as two
+// sgemm_strided_batched per panel on the same (torch) stream.  As
+// (b, n, n) fp32 contiguous symmetric row-major, n % 4 == 0, is
+// overwritten with scaled reflectors (LAPACK lower format when read
+// column-major; D stays on the diagonal).  Returns (E, tau, status);
+// a nonzero status entry means that matrix aborted (bounded-spin
+// barrier or non-finite data) and the caller must fall back.
+std::vector<torch::Tensor> sytrd_batched_custom_(torch::Tensor As) {
+  ensure_init();
+  TORCH_CHECK(As.is_cuda() && As.dim() == 3 && As.size(1) == As.size(2),
+              "sytrd_batched_custom_: (b, n, n) GPU tensor required");
+  TORCH_CHECK(As.scalar_type() == at::kFloat && As.is_contiguous(),
+              "sytrd_batched_custom_: fp32 contiguous required");
+  const int b = (int)As.size(0);
+  const int n = (int)As.size(1);
+  TORCH_CHECK(n % 4 == 0 && n >= 128,
+              "sytrd_batched_custom_: n % 4 == 0 and n >= 128 required");
+  // pick rows-per-workgroup: smallest R (most workgroups = most
+  // bandwidth per matrix) whose grid is guaranteed co-resident --
+  // a non-resident workgroup would deadlock the grid barrier
+  static const int cand[] = {24, 32, 48, 64, 96, 128};
+  int R = 0, wgs = 0, cap = 0;
+  for (int c : cand) {
+    int w = (n + c - 1) / c;
+    int cp = sytrd_panel_max_blocks(c);
+    if (w <= 128 && (long)w * b <= cp) { R = c; wgs = w; cap = cp; break; }
+  }
+  TORCH_CHECK(R > 0, "sytrd_batched_custom_: no resident grid for n=",
+              n, " b=", b, " (cap(128)=", sytrd_panel_max_blocks(128),
+              ") -- fall back to rocSOLVER");
+  (void)cap;
+  const int wgs_alloc = wgs;
+  auto opts = As.options();
+  auto W = at::empty({b, 64L, (long)n}, opts);
+  auto E = at::zeros({b, (long)n}, opts);
+  auto tau = at::zeros({b, (long)n}, opts);
+  auto scratch = at::empty({b, (long)wgs_alloc, 130L}, opts);
+  auto syncc = at::zeros({b}, opts.dtype(at::kInt));
+  auto status = at::zeros({b}, opts.dtype(at::kInt));
+  auto stream = c10::hip::getCurrentHIPStream();
+  ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  const float one = 1.f, neg1 = -1.f;
+  float* Ap = As.data_ptr<float>();
+  float* Wp = W.data_ptr<float>();
+  for (int j0 = 0; j0 < n - 1; j0 += 64) {
+    const int ib = std::min(64, n - 1 - j0);
+    const int wg = (n - j0 + R - 1) / R;
+    sytrd_panels_device(As, W, E, tau, scratch, syncc, status, j0, ib,
+                        R, wgs_alloc, wg);
+    const int t = j0 + ib;
+    const int M = n - t;
+    if (M > 0) {
+      // row-major: A[t:, t:] -= V2^T W2 + W2^T V2 with V2 = A[j0:t, t:]
+      // (reflector rows), W2 = W[0:ib, t:].  In rocBLAS column-major
+      // terms (buffers viewed as M x ib with ld = n):
+      // C -= P Q^T + Q P^T, C symmetric so the transposed view is fine.
+      float* P = Ap + (size_t)j0 * n + t;      // V2, ld n, stride n*n
+      float* Q = Wp + t;                       // W2, ld n, stride 64*n
+      float* C = Ap + (size_t)t * n + t;       // trailing, ld n
+      ROCBLASCHECK(rocblas_sgemm_strided_batched(
+          g_main_handle, rocblas_operation_none,
+          rocblas_operation_transpose, M, M, ib, &neg1, P, n,
+          (rocblas_stride)n * n, Q, n, (rocblas_stride)64 * n, &one, C,
+          n, (rocblas_stride)n * n, b));
+      ROCBLASCHECK(rocblas_sgemm_strided_batched(
+          g_main_handle, rocblas_operation_none,
+          rocblas_operation_transpose, M, M, ib, &neg1, Q, n,
+          (rocblas_stride)64 * n, P, n, (rocblas_stride)n * n, &one, C,
+          n, (rocblas_stride)n * n, b));
+    }
+  }
+  return {E, tau, status};
+}
+
+// Tridiagonal divide-and-conquer eigensolve on a pool slot (for
+// overlapping the per-matrix stedc stage of the custom sytrd path).
+// D (n,) is overwritten with ascending eigenvalues; returns (C, info)
+// with C the eigenvector matrix, column-major in the buffer.
+std::vector<torch::Tensor> stedc_slot_(torch::Tensor D, torch::Tensor E,
+                                       int slot) {
+  ensure_init();
+  const long n = D.numel();
+  TORCH_CHECK(D.is_cuda() && E.is_cuda() && E.numel() >= n - 1 &&
+                  D.is_contiguous() && E.is_contiguous(),
+              "stedc_slot_: contiguous device D/E required");
+  auto C = at::eye(n, D.options());
+  auto info = at::empty({1}, D.options().dtype(at::kInt));
+  auto stream = c10::hip::getCurrentHIPStream();
+  rocblas_handle h = g_main_handle;
+  if (slot >= 0) {
+    Slot& s = g_pool[slot % POOL];
+    HIPCHECK(hipEventRecord(g_acq_event, stream.stream()));
+    HIPCHECK(hipStreamWaitEvent(s.stream, g_acq_event, 0));
+    record_on(D, s.stream);
+    record_on(E, s.stream);
+    record_on(C, s.stream);
+    record_on(info, s.stream);
+    h = s.handle;
+  } else {
+    ROCBLASCHECK(rocblas_set_stream(g_main_handle, stream.stream()));
+  }
+  ROCBLASCHECK(rocsolver_sstedc(h, rocblas_evect_original,
+                                (rocblas_int)n, D.data_ptr<float>(),
+                                E.data_ptr<float>(), C.data_ptr<float>(),
+                                (rocblas_int)n, info.data_ptr<int>()));
+  return {C, info};
+}
+
 std::vector<torch::Tensor> stedc_(torch::Tensor D, torch::Tensor E) {
   ensure_init();
   const long n = D.numel();
@@ -393,8 +511,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("join_pool_", &join_pool_,
         "torch current stream waits on all pool streams");
   m.def("sytrd_", &sytrd_, "in-place tridiagonalization; returns D,E,tau");
+  m.def("sytrd_batched_custom_", &sytrd_batched_custom_,
+        "hand-written batched blocked tridiagonalization (persistent "
+        "panel kernel + strided-batched trailing GEMMs); in-place; "
+        "returns (E, tau, status)");
   m.def("stedc_", &stedc_,
         "tridiagonal D&C eigensolve; returns (C eigvec col-major, info)");
+  m.def("stedc_slot_", &stedc_slot_,
+        "tridiagonal D&C eigensolve on a pool slot (join_pool_ after); "
+        "returns (C, info)", py::arg("D"), py::arg("E"),
+        py::arg("slot") = -1);
   m.def("ormtr_", &ormtr_,
         "C <- Q C with Q from sytrd reflectors (in-place)");
   m.def("syevd_pool_", &syevd_pool_,

@@ -132,6 +132,61 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
 # member) buys another 3x+ of batching.
 JAC_DISPATCH_MAX = 64
 PAD_RATIO = 1.16
+# buckets at or above this (padded) dim go through the hand-written
+# persistent-panel tridiagonalization (sytrd_panel.hip) + pool-stream
+# rocSOLVER stedc + the batched WY back-transform below, instead of
+# rocSOLVER's syevd whose latrd panel storm is latency-bound.
+# KFAC_CUSTOM_SYTRD=0 disables; KFAC_SYTRD_MIN overrides the cutoff.
+SYTRD_DISPATCH_MIN = 1500
+
+
+def _custom_sytrd_on() -> bool:
+    if os.environ.get("KFAC_CUSTOM_SYTRD", "1") == "0":
+        return False
+    from kfac_pytorch_amd.ops import _ext
+    if not _ext.has_solver():
+        return False
+    return hasattr(_ext.load_solver(), "sytrd_batched_custom_")
+
+
+def _wy_backtransform(stacked: torch.Tensor, tauT: torch.Tensor,
+                      Cstack: torch.Tensor) -> torch.Tensor:
+    """Back-transform tridiagonal eigenvectors by the factored Q from
+    the custom sytrd: returns the (b, n, n) eigenvector tensor with
+    COLUMNS = eigenvectors (the eigh contract).
+
+    ``stacked`` holds the scaled reflectors (row j: unit at j+1, v at
+    j+2..; junk below); ``Cstack`` is the raw stedc output buffer
+    (eigenvectors column-major).  Q = B_0 B_1 ... B_{P-1} with each
+    64-wide block in compact WY form I - V T V^T, where
+    T^{-1} = diag(1/tau) + strict_upper(V^T V) -- so each block apply
+    is three strided-batched GEMMs plus one small batched triangular
+    solve, all MFMA-backed, replacing rocSOLVER's sormtr (which runs
+    the same math as small serial larft/larfb kernels).
+    """
+    b, n, _ = stacked.shape
+    res = Cstack.mT.contiguous()  # semantic Z (columns = eigvecs of T)
+    dev = stacked.device
+    col = torch.arange(n, device=dev)
+    for p in reversed(range(0, n - 1, 64)):
+        jb = min(64, (n - 1) - p)
+        tau_p = tauT[:, p:p + jb]
+        # V^T as rows, columns < unit zeroed, unit explicit in storage
+        mask = (col[p:].unsqueeze(0)
+                >= (p + 1 + torch.arange(jb, device=dev)).unsqueeze(1))
+        Vt = stacked[:, p:p + jb, p:] * mask
+        zr = tau_p == 0
+        if bool(zr.any()):
+            # degenerate reflectors (H = I): zero v, tau -> 1 keeps
+            # T well-defined and the block exact
+            Vt = Vt * (~zr).unsqueeze(-1)
+            tau_p = torch.where(zr, torch.ones_like(tau_p), tau_p)
+        S = torch.bmm(Vt, Vt.mT)
+        Tinv = S.triu(1) + torch.diag_embed(1.0 / tau_p)
+        Y = torch.bmm(Vt, res[:, p:, :])
+        TY = torch.linalg.solve_triangular(Tinv, Y, upper=True)
+        res[:, p:, :] -= torch.bmm(Vt.mT, TY)
+    return res
 
 
 def _pad_buckets(dims):
@@ -202,7 +257,54 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
         slot = 0
         issued_on_pool = False
         device = mats[rest[0]].device
+        sytrd_min = int(os.environ.get("KFAC_SYTRD_MIN",
+                                       SYTRD_DISPATCH_MIN))
+        use_custom = _custom_sytrd_on()
+        custom_jobs = []  # (members, n4, stacked, tauT, D, Cs)
         for n, members in _pad_buckets(dims):
+            if use_custom and n >= sytrd_min:
+                # custom tier: persistent-panel sytrd on the torch
+                # stream (buckets pipeline: this bucket's stedc on pool
+                # slots overlaps the next bucket's sytrd), WY
+                # back-transform after the single pool join
+                n4 = (n + 3) & ~3  # kernel wants n % 4 == 0
+                b = len(members)
+                stacked = torch.full((b, n4, n4), 0.0, device=device)
+                for k, (m, i) in enumerate(members):
+                    pad = n4 - m
+                    stacked[k, pad:, pad:] = mats[i]
+                    if pad:
+                        stacked[k].diagonal()[:pad] = -1.0
+                try:
+                    E, tauT, status = solver.sytrd_batched_custom_(
+                        stacked)
+                except RuntimeError:
+                    # no resident grid for this (n, b): library tier
+                    try:
+                        W, _info = solver.syevd_batched_(stacked, slot)
+                    except RuntimeError:
+                        singles.extend(i for _, i in members)
+                        continue
+                    _defer_info(_info)
+                    slot += 1
+                    issued_on_pool = True
+                    for k, (m, i) in enumerate(members):
+                        pad = n4 - m
+                        Q = stacked[k].mT
+                        out[i] = (W[k, pad:],
+                                  Q[pad:, pad:] if pad else Q)
+                    continue
+                _defer_info(status)
+                D = stacked.diagonal(dim1=1, dim2=2).contiguous()
+                Cs = []
+                for k in range(b):
+                    C, cinfo = solver.stedc_slot_(D[k], E[k], slot % 8)
+                    _defer_info(cinfo)
+                    slot += 1
+                    Cs.append(C)
+                issued_on_pool = True
+                custom_jobs.append((members, n4, stacked, tauT, D, Cs))
+                continue
             if len(members) < 2:
                 singles.extend(i for _, i in members)
                 continue
@@ -252,6 +354,14 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                     out[i] = mat_eig(mats[i], method="eigh")
         elif issued_on_pool:
             solver.join_pool_()
+        # WY back-transform of the custom-sytrd buckets (torch stream,
+        # after the pool join so every stedc has finished)
+        for members, n4, stacked, tauT, D, Cs in custom_jobs:
+            res = _wy_backtransform(stacked, tauT, torch.stack(Cs))
+            for k, (m, i) in enumerate(members):
+                pad = n4 - m
+                out[i] = (D[k, pad:],
+                          res[k, pad:, pad:] if pad else res[k])
         if os.environ.get("KFAC_EIG_DEBUG"):
             torch.cuda.synchronize()
             import time as _t

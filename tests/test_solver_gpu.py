@@ -162,3 +162,72 @@ def test_mat_inv_multi_padded_batched(solver):
         ref = torch.linalg.inv(damped)
         err = ((inv - ref).norm() / ref.norm()).item()
         assert err < 1e-3, (a.shape, err)
+
+
+def test_custom_sytrd_direct(solver):
+    """Hand-written batched tridiagonalization vs torch.linalg.eigvalsh:
+    the tridiagonal T must be orthogonally similar to A (eigenvalues
+    match) and no matrix may abort."""
+    if not hasattr(solver, "sytrd_batched_custom_"):
+        pytest.skip("old _kfac_solver build")
+    mats = torch.stack([spd(1152, seed=11 + i) for i in range(3)])
+    work = mats.clone()
+    E, tau, status = solver.sytrd_batched_custom_(work)
+    torch.cuda.synchronize()
+    assert int(status.abs().sum()) == 0, status.tolist()
+    for k in range(mats.shape[0]):
+        n = mats.shape[-1]
+        D = work[k].diagonal()
+        T = (torch.diag(D) + torch.diag(E[k][:n - 1], 1)
+             + torch.diag(E[k][:n - 1], -1))
+        ev = torch.linalg.eigvalsh(T)
+        ev_ref = torch.linalg.eigvalsh(mats[k])
+        scale = float(ev_ref.abs().max())
+        err = float((ev - ev_ref).abs().max()) / scale
+        assert err < 5e-5, (k, err)
+
+
+def test_custom_sytrd_full_path(solver, monkeypatch):
+    """Full custom tier through mat_eig_multi (sytrd -> stedc -> WY
+    back-transform), forced on for every bucket size, with padded
+    buckets and rank-deficient members -- vs the eigh oracle."""
+    if not hasattr(solver, "sytrd_batched_custom_"):
+        pytest.skip("old _kfac_solver build")
+    from kfac_pytorch_amd.ops.linalg import mat_eig_multi
+    monkeypatch.setenv("KFAC_SYTRD_MIN", "1000")
+    monkeypatch.setenv("KFAC_CUSTOM_SYTRD", "1")
+    # 2049/2304 pad into one bucket; 1397 tests the n%4 rounding;
+    # the rank-deficient member mimics early-training conv factors
+    dims = [2304, 2049, 2048, 1397, 1152, 1152]
+    mats = [spd(m, seed=m) for m in dims]
+    lowrank = torch.randn(1152, 200, device="cuda")
+    mats[-1] = lowrank @ lowrank.t() / 200
+    out = mat_eig_multi(mats, need_sorted=False)
+    torch.cuda.synchronize()
+    for a, (w, Q) in zip(mats, out):
+        assert w.shape[0] == a.shape[0]
+        assert Q.shape == a.shape
+        check_eig(a, w, Q)
+        w_ref = torch.linalg.eigvalsh(a)
+        torch.testing.assert_close(torch.sort(w).values, w_ref,
+                                   rtol=2e-3, atol=2e-3)
+
+
+def test_custom_sytrd_precondition_accuracy(solver, monkeypatch):
+    """What K-FAC actually consumes: the damped-inverse built from the
+    custom path's eigenpairs must match the exact eigh's."""
+    if not hasattr(solver, "sytrd_batched_custom_"):
+        pytest.skip("old _kfac_solver build")
+    from kfac_pytorch_amd.ops.linalg import mat_eig_multi
+    monkeypatch.setenv("KFAC_SYTRD_MIN", "1000")
+    a = spd(1536, seed=5)
+    (res,) = [mat_eig_multi([a], need_sorted=False)[0]]
+    torch.cuda.synchronize()
+    w, Q = res
+    damping = 0.002
+    P = Q @ torch.diag(1.0 / (w.clamp_min(0) + damping)) @ Q.mT
+    w_ref, Q_ref = torch.linalg.eigh(a)
+    P_ref = (Q_ref @ torch.diag(1.0 / (w_ref.clamp_min(0) + damping))
+             @ Q_ref.mT)
+    err = ((P - P_ref).norm() / P_ref.norm()).item()
+    assert err < 5e-3, err
