@@ -31,6 +31,7 @@
 #include <c10/hip/HIPCachingAllocator.h>
 
 #include <algorithm>
+#include <cstdlib>
 #include <sstream>
 #include <vector>
 
@@ -377,20 +378,34 @@ std::vector<torch::Tensor> sytrd_batched_custom_(torch::Tensor As) {
   const int n = (int)As.size(1);
   TORCH_CHECK(n % 4 == 0 && n >= 128,
               "sytrd_batched_custom_: n % 4 == 0 and n >= 128 required");
-  // pick rows-per-workgroup: smallest R (most workgroups = most
-  // bandwidth per matrix) whose grid is guaranteed co-resident --
-  // a non-resident workgroup would deadlock the grid barrier
-  static const int cand[] = {24, 32, 48, 64, 96, 128};
-  int R = 0, wgs = 0, cap = 0;
+  // pick rows-per-workgroup: aim for ~512 TOTAL workgroups across the
+  // batch (fills the chip's bandwidth) but no more -- the per-column
+  // cross-workgroup reductions cost O(wgs^2) scratch reads per matrix
+  // and every extra workgroup is another barrier poller, both measured
+  // dominant when wgs_per_matrix was maximized.  The grid must also be
+  // guaranteed co-resident (a non-resident workgroup would deadlock
+  // the grid barrier).
+  static const int cand[] = {24, 32, 36, 48, 64, 96, 128};
+  int total_target = 512;
+  if (const char* e = getenv("KFAC_SYTRD_WGS"))
+    total_target = std::max(64, atoi(e));
+  const int want = std::min(128, std::max(1, total_target / b));
+  int R = 0, wgs = 0;
   for (int c : cand) {
     int w = (n + c - 1) / c;
+    if (w > want) continue;
     int cp = sytrd_panel_max_blocks(c);
-    if (w <= 128 && (long)w * b <= cp) { R = c; wgs = w; cap = cp; break; }
+    if ((long)w * b <= cp) { R = c; wgs = w; break; }
+  }
+  if (R == 0) {
+    // batch too large for the target: take the coarsest grid that fits
+    int w = (n + 127) / 128;
+    int cp = sytrd_panel_max_blocks(128);
+    if ((long)w * b <= cp) { R = 128; wgs = w; }
   }
   TORCH_CHECK(R > 0, "sytrd_batched_custom_: no resident grid for n=",
               n, " b=", b, " (cap(128)=", sytrd_panel_max_blocks(128),
               ") -- fall back to rocSOLVER");
-  (void)cap;
   const int wgs_alloc = wgs;
   auto opts = As.options();
   auto W = at::empty({b, 64L, (long)n}, opts);

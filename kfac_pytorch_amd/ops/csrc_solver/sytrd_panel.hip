@@ -71,8 +71,15 @@ __device__ bool grid_barrier(unsigned* cnt, int* status, int wgs,
     long spins = 0;
     while (__hip_atomic_load(cnt, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT) < target) {
-      __builtin_amdgcn_s_sleep(2);
-      if ((++spins & 4095) == 0) {
+      // escalating backoff: hot polls only for the first ~2 us, then
+      // ~1.7 us sleeps -- hundreds of concurrent pollers on a few
+      // counter words would otherwise steal chip bandwidth from the
+      // matvec streams (guide: polling-cost row)
+      if (spins < 32)
+        __builtin_amdgcn_s_sleep(4);
+      else
+        __builtin_amdgcn_s_sleep(64);
+      if ((++spins & 255) == 0) {
         if (__hip_atomic_load(status, __ATOMIC_RELAXED,
                               __HIP_MEMORY_SCOPE_AGENT) != 0) break;
         if (spins > SPIN_CAP) {
@@ -162,9 +169,14 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         slabV[(i - 1) * RS + (r - r0)] = v;
         slabW[(i - 1) * RS + (r - r0)] = Wprev[r];
       }
-      // per-column correction coefficients (one wave, lane = c)
+      // per-column correction coefficients (one wave, lane = c).
+      // c == i-1's V value at row j IS the unit element (j = (j-1)+1)
+      // and its stored copy is being written by another workgroup in
+      // THIS phase -- use the exact 1.0, never the racy load.
       if (wave == 0 && lane < i) {
-        float cb = A_m[(size_t)(j0 + lane) * n + j];
+        float cb = (lane == i - 1)
+                       ? 1.f
+                       : A_m[(size_t)(j0 + lane) * n + j];
         coefA[lane] = W_m[(size_t)lane * n + j] + 2.f * alpha[lane] * cb;
         coefB[lane] = cb;
       }
@@ -281,14 +293,19 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         if (lane == 0)
           for (; t < n && (t & 3); ++t) dot += Ar[t] * Aj[t];
         t = (j + 2 + 3) & ~3;
+        // four independent accumulator chains: a single serial fmaf
+        // chain caps each wave at ~2 B/cycle and starves the HBM
+        // stream
+        float4 acc = {0.f, 0.f, 0.f, 0.f};
         for (int tb = t + 4 * lane; tb < n; tb += 256) {
           const float4 a = *(const float4*)(Ar + tb);
           const float4 xv = *(const float4*)(Aj + tb);
-          dot = fmaf(a.x, xv.x, dot);
-          dot = fmaf(a.y, xv.y, dot);
-          dot = fmaf(a.z, xv.z, dot);
-          dot = fmaf(a.w, xv.w, dot);
+          acc.x = fmaf(a.x, xv.x, acc.x);
+          acc.y = fmaf(a.y, xv.y, acc.y);
+          acc.z = fmaf(a.z, xv.z, acc.z);
+          acc.w = fmaf(a.w, xv.w, acc.w);
         }
+        dot += (acc.x + acc.y) + (acc.z + acc.w);
         // fold: w_pre = A[r][j+1] + s*dot - corrections (lane = c)
         float contrib = s * dot;
         const int rl = r - r0;
