@@ -357,7 +357,7 @@ std::vector<torch::Tensor> sytrd_panels_device(
     torch::Tensor As, torch::Tensor W, torch::Tensor E,
     torch::Tensor tau, torch::Tensor scratch, torch::Tensor syncc,
     torch::Tensor status, int j0, int ib, int R, int wgs_alloc, int wgs);
-int sytrd_panel_max_blocks(int R);
+int sytrd_panel_max_blocks(int R, int xlen_max);
 
 // Custom batched blocked tridiagonalization (sytrd_panel.hip): the
 // hand-written persistent-panel kernel replaces rocSOLVER's latrd
@@ -386,25 +386,25 @@ std::vector<torch::Tensor> sytrd_batched_custom_(torch::Tensor As) {
   // guaranteed co-resident (a non-resident workgroup would deadlock
   // the grid barrier).
   static const int cand[] = {24, 32, 36, 48, 64, 96, 128};
-  int total_target = 512;
+  int total_target = 768;
   if (const char* e = getenv("KFAC_SYTRD_WGS"))
     total_target = std::max(64, atoi(e));
-  const int want = std::min(128, std::max(1, total_target / b));
+  const int want = std::min(192, std::max(1, total_target / b));
   int R = 0, wgs = 0;
   for (int c : cand) {
     int w = (n + c - 1) / c;
     if (w > want) continue;
-    int cp = sytrd_panel_max_blocks(c);
+    int cp = sytrd_panel_max_blocks(c, n);
     if ((long)w * b <= cp) { R = c; wgs = w; break; }
   }
   if (R == 0) {
     // batch too large for the target: take the coarsest grid that fits
     int w = (n + 127) / 128;
-    int cp = sytrd_panel_max_blocks(128);
+    int cp = sytrd_panel_max_blocks(128, n);
     if ((long)w * b <= cp) { R = 128; wgs = w; }
   }
   TORCH_CHECK(R > 0, "sytrd_batched_custom_: no resident grid for n=",
-              n, " b=", b, " (cap(128)=", sytrd_panel_max_blocks(128),
+              n, " b=", b, " (cap(128)=", sytrd_panel_max_blocks(128, n),
               ") -- fall back to rocSOLVER");
   const int wgs_alloc = wgs;
   auto opts = As.options();

@@ -137,6 +137,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
   float* qB = qA + SYTRD_NB;                 // [SYTRD_NB]
   float* alpha = qB + SYTRD_NB;              // [SYTRD_NB]
   float* red = alpha + SYTRD_NB;             // [4] cross-wave scratch
+  // the pre-scale x row, staged once per column (16B-aligned carve)
+  float* x_lds = (float*)((((size_t)(red + 4)) + 15) & ~(size_t)15);
 
   if (__hip_atomic_load(st, __ATOMIC_RELAXED,
                         __HIP_MEMORY_SCOPE_AGENT) != 0)
@@ -152,7 +154,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       // alpha of the previous column from the pwv partials
       float p = 0.f;
       for (int t = tid; t < wgs; t += 256)
-        p += scr[(size_t)t * SYTRD_SLOTS + 129];
+        p += scr[(size_t)129 * wgs_alloc + t];
       p = wave_reduce(p);
       if (lane == 0) red[wave] = p;
       __syncthreads();
@@ -216,7 +218,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         if (!isfinite(t0))
           __hip_atomic_store(st, 2, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT);
-        scr[(size_t)w * SYTRD_SLOTS + 0] = t0;
+        scr[(size_t)0 * wgs_alloc + w] = t0;
       }
       // pV/pW: waves split the c-range, lanes... lanes = c, loop rows.
       const int rlo = max(r0, j + 1);
@@ -230,8 +232,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         pv = wave_reduce(pv);
         pw = wave_reduce(pw);
         if (lane == 0) {
-          scr[(size_t)w * SYTRD_SLOTS + 1 + c] = pv;
-          scr[(size_t)w * SYTRD_SLOTS + 65 + c] = pw;
+          scr[(size_t)(1 + c) * wgs_alloc + w] = pv;
+          scr[(size_t)(65 + c) * wgs_alloc + w] = pw;
         }
       }
     }
@@ -246,7 +248,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         if (live) {
           float acc = 0.f;
           for (int t = 0; t < wgs; ++t)
-            acc += scr[(size_t)t * SYTRD_SLOTS + s];
+            acc += scr[(size_t)s * wgs_alloc + t];
           sums[s] = acc;
         }
       }
@@ -279,10 +281,17 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       qA[lane] = sW + 2.f * alpha[lane] * sV;
       qB[lane] = sV;
     }
-    __syncthreads();
-    // the trailing matvec: one wave per row, float4, pre-scale x
+    // stage the pre-scale x row into LDS once per column: the matvec
+    // would otherwise re-read it from L2 per row, doubling the VMEM
+    // issue in the hot loop (per-CU streaming is issue-bound)
     {
       const float* Aj = A_m + (size_t)j * n;
+      for (int t4 = j0 + 4 * tid; t4 < n; t4 += 1024)
+        *(float4*)(x_lds + (t4 - j0)) = *(const float4*)(Aj + t4);
+    }
+    __syncthreads();
+    // the trailing matvec: one wave per row, float4, x from LDS
+    {
       const int rlo = max(r0, j + 1);
       float pwv_acc = 0.f;
       for (int r = rlo + wave; r < r1; r += 4) {
@@ -291,15 +300,17 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         int t = j + 2;
         // head to 16B alignment (n % 4 == 0 -> row bases aligned)
         if (lane == 0)
-          for (; t < n && (t & 3); ++t) dot += Ar[t] * Aj[t];
+          for (; t < n && (t & 3); ++t)
+            dot += Ar[t] * x_lds[t - j0];
         t = (j + 2 + 3) & ~3;
         // four independent accumulator chains: a single serial fmaf
         // chain caps each wave at ~2 B/cycle and starves the HBM
         // stream
         float4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 2
         for (int tb = t + 4 * lane; tb < n; tb += 256) {
           const float4 a = *(const float4*)(Ar + tb);
-          const float4 xv = *(const float4*)(Aj + tb);
+          const float4 xv = *(const float4*)(x_lds + (tb - j0));
           acc.x = fmaf(a.x, xv.x, acc.x);
           acc.y = fmaf(a.y, xv.y, acc.y);
           acc.z = fmaf(a.z, xv.z, acc.z);
@@ -325,7 +336,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       if (lane == 0) red[wave] = pwv_acc;
       __syncthreads();
       if (tid == 0)
-        scr[(size_t)w * SYTRD_SLOTS + 129] =
+        scr[(size_t)129 * wgs_alloc + w] =
             red[0] + red[1] + red[2] + red[3];
     }
     tau_prev = tau_j;
@@ -337,7 +348,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
   {
     float p = 0.f;
     for (int t = tid; t < wgs; t += 256)
-      p += scr[(size_t)t * SYTRD_SLOTS + 129];
+      p += scr[(size_t)129 * wgs_alloc + t];
     p = wave_reduce(p);
     if (lane == 0) red[wave] = p;
     __syncthreads();
@@ -388,7 +399,8 @@ std::vector<torch::Tensor> sytrd_panels_device(
   auto stream = c10::hip::getCurrentHIPStream();
   const int RS = R | 1;
   const size_t lds = sizeof(float) *
-      (2 * SYTRD_NB * RS + RS + SYTRD_SLOTS + 5 * SYTRD_NB + 4);
+      (2 * SYTRD_NB * RS + RS + SYTRD_SLOTS + 5 * SYTRD_NB + 4 +
+       (size_t)(n - j0)) + 16;
   hipError_t err = hipMemsetAsync(syncc.data_ptr(), 0,
                                   sizeof(unsigned) * b, stream.stream());
   TORCH_CHECK(err == hipSuccess, "sytrd memset: ",
@@ -404,10 +416,11 @@ std::vector<torch::Tensor> sytrd_panels_device(
   return {};
 }
 
-int sytrd_panel_max_blocks(int R) {
+int sytrd_panel_max_blocks(int R, int xlen_max) {
   const int RS = R | 1;
   const size_t lds = sizeof(float) *
-      (2 * SYTRD_NB * RS + RS + SYTRD_SLOTS + 5 * SYTRD_NB + 4);
+      (2 * SYTRD_NB * RS + RS + SYTRD_SLOTS + 5 * SYTRD_NB + 4 +
+       (size_t)xlen_max) + 16;
   int nb = 0;
   hipError_t err = hipOccupancyMaxActiveBlocksPerMultiprocessor(
       &nb, (const void*)latrd_panel_kernel, 256, lds);
