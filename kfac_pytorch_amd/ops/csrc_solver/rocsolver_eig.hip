@@ -356,7 +356,8 @@ std::vector<torch::Tensor> sytrd_(torch::Tensor A) {
 std::vector<torch::Tensor> sytrd_panels_device(
     torch::Tensor As, torch::Tensor W, torch::Tensor E,
     torch::Tensor tau, torch::Tensor scratch, torch::Tensor syncc,
-    torch::Tensor status, int j0, int ib, int R, int wgs_alloc, int wgs);
+    torch::Tensor status, int j0, int ib, int R, int wgs_alloc, int wgs,
+    int mode);
 int sytrd_panel_max_blocks(int R, int xlen_max);
 
 // Custom batched blocked tridiagonalization (sytrd_panel.hip): the
@@ -406,6 +407,8 @@ std::vector<torch::Tensor> sytrd_batched_custom_(torch::Tensor As) {
   TORCH_CHECK(R > 0, "sytrd_batched_custom_: no resident grid for n=",
               n, " b=", b, " (cap(128)=", sytrd_panel_max_blocks(128, n),
               ") -- fall back to rocSOLVER");
+  int mode = 0;  // ablation: 1 = skip matvec dot, 2 = barriers only
+  if (const char* e = getenv("KFAC_SYTRD_MODE")) mode = atoi(e);
   const int wgs_alloc = wgs;
   auto opts = As.options();
   auto W = at::empty({b, 64L, (long)n}, opts);
@@ -423,7 +426,7 @@ std::vector<torch::Tensor> sytrd_batched_custom_(torch::Tensor As) {
     const int ib = std::min(64, n - 1 - j0);
     const int wg = (n - j0 + R - 1) / R;
     sytrd_panels_device(As, W, E, tau, scratch, syncc, status, j0, ib,
-                        R, wgs_alloc, wg);
+                        R, wgs_alloc, wg, mode);
     const int t = j0 + ib;
     const int M = n - t;
     if (M > 0) {

@@ -110,7 +110,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
     float* __restrict__ E, float* __restrict__ tau,
     float* __restrict__ scratch, unsigned* __restrict__ syncc,
     int* __restrict__ status, int n, int j0, int ib, int R,
-    int wgs_alloc) {
+    int wgs_alloc, int mode) {
   const int bm = blockIdx.y;
   const int w = blockIdx.x;
   const int wgs = gridDim.x;
@@ -149,6 +149,11 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
 
   for (int i = 0; i < ib; ++i) {
     const int j = j0 + i;
+    if (mode & 2) {  // ablation: barriers only
+      if (!grid_barrier(cnt, st, wgs, ++phase)) return;
+      if (!grid_barrier(cnt, st, wgs, ++phase)) return;
+      continue;
+    }
     // ================= phase A =================
     if (i > 0) {
       // alpha of the previous column from the pwv partials
@@ -308,7 +313,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         // stream
         float4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 2
-        for (int tb = t + 4 * lane; tb < n; tb += 256) {
+        for (int tb = t + 4 * lane; (mode & 1) == 0 && tb < n;
+             tb += 256) {
           const float4 a = *(const float4*)(Ar + tb);
           const float4 xv = *(const float4*)(x_lds + (tb - j0));
           acc.x = fmaf(a.x, xv.x, acc.x);
@@ -393,7 +399,7 @@ std::vector<torch::Tensor> sytrd_panels_device(
     torch::Tensor As, torch::Tensor W, torch::Tensor E,
     torch::Tensor tau, torch::Tensor scratch, torch::Tensor syncc,
     torch::Tensor status, int j0, int ib, int R, int wgs_alloc,
-    int wgs) {
+    int wgs, int mode) {
   const int n = (int)As.size(1);
   const int b = (int)As.size(0);
   auto stream = c10::hip::getCurrentHIPStream();
@@ -409,7 +415,7 @@ std::vector<torch::Tensor> sytrd_panels_device(
       As.data_ptr<float>(), W.data_ptr<float>(), E.data_ptr<float>(),
       tau.data_ptr<float>(), scratch.data_ptr<float>(),
       (unsigned*)syncc.data_ptr(), status.data_ptr<int>(), n, j0, ib, R,
-      wgs_alloc);
+      wgs_alloc, mode);
   err = hipGetLastError();
   TORCH_CHECK(err == hipSuccess, "latrd_panel_kernel launch: ",
               hipGetErrorString(err));
