@@ -387,7 +387,7 @@ std::vector<torch::Tensor> sytrd_batched_custom_(torch::Tensor As) {
   // guaranteed co-resident (a non-resident workgroup would deadlock
   // the grid barrier).
   static const int cand[] = {24, 32, 36, 48, 64, 96, 128};
-  int total_target = 768;
+  int total_target = 384;
   if (const char* e = getenv("KFAC_SYTRD_WGS"))
     total_target = std::max(64, atoi(e));
   const int want = std::min(192, std::max(1, total_target / b));

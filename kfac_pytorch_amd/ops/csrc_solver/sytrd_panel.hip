@@ -295,47 +295,69 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         *(float4*)(x_lds + (t4 - j0)) = *(const float4*)(Aj + t4);
     }
     __syncthreads();
-    // the trailing matvec: one wave per row, float4, x from LDS
+    // the trailing matvec: each wave streams FOUR rows concurrently
+    // (independent load/acc chains -- a single row's loads serialize
+    // behind its reduction and pay full HBM latency per row), x from
+    // LDS shared across the group
     {
       const int rlo = max(r0, j + 1);
       float pwv_acc = 0.f;
-      for (int r = rlo + wave; r < r1; r += 4) {
-        const float* Ar = A_m + (size_t)r * n;
-        float dot = 0.f;
+      for (int rg = rlo + 4 * wave; rg < r1; rg += 16) {
+        const int nr = min(4, r1 - rg);
+        // rows rg..rg+nr-1; k >= nr alias row rg (loads valid,
+        // results discarded)
+        const float* Ar[4];
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+          Ar[k] = A_m + (size_t)(k < nr ? rg + k : rg) * n;
+        float dot[4] = {0.f, 0.f, 0.f, 0.f};
         int t = j + 2;
         // head to 16B alignment (n % 4 == 0 -> row bases aligned)
         if (lane == 0)
-          for (; t < n && (t & 3); ++t)
-            dot += Ar[t] * x_lds[t - j0];
+          for (; t < n && (t & 3); ++t) {
+            const float xh = x_lds[t - j0];
+#pragma unroll
+            for (int k = 0; k < 4; ++k) dot[k] += Ar[k][t] * xh;
+          }
         t = (j + 2 + 3) & ~3;
-        // four independent accumulator chains: a single serial fmaf
-        // chain caps each wave at ~2 B/cycle and starves the HBM
-        // stream
-        float4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll 2
-        for (int tb = t + 4 * lane; (mode & 1) == 0 && tb < n;
-             tb += 256) {
-          const float4 a = *(const float4*)(Ar + tb);
-          const float4 xv = *(const float4*)(x_lds + (tb - j0));
-          acc.x = fmaf(a.x, xv.x, acc.x);
-          acc.y = fmaf(a.y, xv.y, acc.y);
-          acc.z = fmaf(a.z, xv.z, acc.z);
-          acc.w = fmaf(a.w, xv.w, acc.w);
+        if ((mode & 1) == 0) {
+          float4 acc[4];
+#pragma unroll
+          for (int k = 0; k < 4; ++k)
+            acc[k] = {0.f, 0.f, 0.f, 0.f};
+          for (int tb = t + 4 * lane; tb < n; tb += 256) {
+            const float4 xv = *(const float4*)(x_lds + (tb - j0));
+#pragma unroll
+            for (int k = 0; k < 4; ++k) {
+              const float4 a = *(const float4*)(Ar[k] + tb);
+              acc[k].x = fmaf(a.x, xv.x, acc[k].x);
+              acc[k].y = fmaf(a.y, xv.y, acc[k].y);
+              acc[k].z = fmaf(a.z, xv.z, acc[k].z);
+              acc[k].w = fmaf(a.w, xv.w, acc[k].w);
+            }
+          }
+#pragma unroll
+          for (int k = 0; k < 4; ++k)
+            dot[k] += (acc[k].x + acc[k].y) + (acc[k].z + acc[k].w);
         }
-        dot += (acc.x + acc.y) + (acc.z + acc.w);
-        // fold: w_pre = A[r][j+1] + s*dot - corrections (lane = c)
-        float contrib = s * dot;
-        const int rl = r - r0;
-        if (lane < i)
-          contrib -= slabV[lane * RS + rl] * qA[lane] +
-                     slabW[lane * RS + rl] * qB[lane];
-        if (lane == 0) contrib += Ar[j + 1];
-        contrib = wave_reduce(contrib);
-        if (lane == 0) {
-          float w1 = tau_j * contrib;
-          W_m[(size_t)i * n + r] = w1;
-          float vt = (r == j + 1) ? 1.f : s * x_own[rl];
-          pwv_acc += w1 * vt;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+          if (k >= nr) break;
+          const int r = rg + k;
+          // fold: w_pre = A[r][j+1] + s*dot - corrections (lane = c)
+          float contrib = s * dot[k];
+          const int rl = r - r0;
+          if (lane < i)
+            contrib -= slabV[lane * RS + rl] * qA[lane] +
+                       slabW[lane * RS + rl] * qB[lane];
+          if (lane == 0) contrib += Ar[k][j + 1];
+          contrib = wave_reduce(contrib);
+          if (lane == 0) {
+            float w1 = tau_j * contrib;
+            W_m[(size_t)i * n + r] = w1;
+            float vt = (r == j + 1) ? 1.f : s * x_own[rl];
+            pwv_acc += w1 * vt;
+          }
         }
       }
       pwv_acc = wave_reduce(pwv_acc);  // lane0-only values; harmless
