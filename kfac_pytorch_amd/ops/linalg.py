@@ -132,16 +132,23 @@ def mat_eig(x: torch.Tensor, method: str = "auto"
 # member) buys another 3x+ of batching.
 JAC_DISPATCH_MAX = 64
 PAD_RATIO = 1.16
-# buckets at or above this (padded) dim go through the hand-written
-# persistent-panel tridiagonalization (sytrd_panel.hip) + pool-stream
-# rocSOLVER stedc + the batched WY back-transform below, instead of
-# rocSOLVER's syevd whose latrd panel storm is latency-bound.
-# KFAC_CUSTOM_SYTRD=0 disables; KFAC_SYTRD_MIN overrides the cutoff.
+# KFAC_CUSTOM_SYTRD=1 routes buckets at or above KFAC_SYTRD_MIN
+# (default 1500, padded dim) through the hand-written persistent-panel
+# tridiagonalization (sytrd_panel.hip) + pool-stream rocSOLVER stedc +
+# the batched WY back-transform below.  OFF by default: measured on
+# MI355X (profiles/PERFORMANCE.md round 2) the custom path reaches
+# parity-minus with rocSOLVER's batched syevd (4608 x 3: 247 ms custom
+# sytrd vs 244 ms for the library's ENTIRE solve) -- both sit on the
+# same per-column tridiagonalization critical path, and the grid
+# barriers + cross-workgroup reductions that a persistent kernel needs
+# cost about what the library's kernel-launch storm does.  Kept as a
+# fully tested opt-in (and the measured study that closes the
+# "replace the latrd storm" roadmap item).
 SYTRD_DISPATCH_MIN = 1500
 
 
 def _custom_sytrd_on() -> bool:
-    if os.environ.get("KFAC_CUSTOM_SYTRD", "1") == "0":
+    if os.environ.get("KFAC_CUSTOM_SYTRD", "0") != "1":
         return False
     from kfac_pytorch_amd.ops import _ext
     if not _ext.has_solver():

@@ -41,9 +41,23 @@ class EigenComputeMixin:
 
     Replaces the reference's serial per-layer eigh loop
     (reference: kfac_preconditioner_eigen.py:98-119).
+
+    ``KFAC_WARM_EIG=1`` routes factors >= KFAC_WARM_MIN (default 512)
+    through the warm-started blocked Jacobi tier (ops/block_jacobi.py)
+    carrying each factor's eigenbasis across updates: in steady-state
+    training (slow factor drift) a couple of batched-GEMM passes
+    replace the cold eigensolve.  The dispatch gates on the MEASURED
+    residual -- a factor whose basis genuinely rotated falls back to
+    the library tier that update (and re-anchors exactly every
+    KFAC_WARM_REANCHOR updates to bound fp32 drift), so accuracy never
+    depends on the steadiness assumption.  Off by default: at
+    fac/kfac_update_freq=1 with fresh sampling noise every step the
+    warm tier only breaks even (measured; see
+    profiles/PERFORMANCE.md round 2).
     """
 
     def _eigendecompose_owned(self):
+        import os
         rank = self.comm.rank()
         work = []
         for m in self.modules:
@@ -56,7 +70,11 @@ class EigenComputeMixin:
             return
         mats = [self.m_A[mod] if kind == "A" else self.m_G[mod]
                 for mod, kind in work]
-        results = mat_eig_multi(mats, need_sorted=False)
+        if (os.environ.get("KFAC_WARM_EIG", "0") == "1"
+                and mats[0].is_cuda):
+            results = self._warm_eig(work, mats)
+        else:
+            results = mat_eig_multi(mats, need_sorted=False)
 
         for (mod, kind), (d, Q) in zip(work, results):
             clamped = d * (d > self.eps)
@@ -66,6 +84,60 @@ class EigenComputeMixin:
             else:
                 self.m_QG[mod].copy_(Q)
                 self.m_dG[mod].copy_(clamped)
+
+    def _warm_eig(self, work, mats):
+        """Warm-tier dispatch: carried-basis blocked Jacobi for the big
+        factors, measured-residual gated, library tier for the rest."""
+        import os
+        import torch as _torch
+        from kfac_pytorch_amd.ops.block_jacobi import \
+            block_jacobi_eigh_batched
+        warm_min = int(os.environ.get("KFAC_WARM_MIN", "512"))
+        reanchor = int(os.environ.get("KFAC_WARM_REANCHOR", "50"))
+        tol = float(os.environ.get("KFAC_WARM_TOL", "5e-6"))
+        max_iters = int(os.environ.get("KFAC_WARM_MAX_ITERS", "12"))
+        if not hasattr(self, "_warm_state"):
+            self._warm_state = {}  # (id(mod), kind) -> [V, age]
+        results = [None] * len(mats)
+        groups = {}
+        for i, (mod, kind) in enumerate(work):
+            key = (id(mod), kind)
+            st = self._warm_state.get(key)
+            n = int(mats[i].shape[-1])
+            # re-anchor exactly (library solve) on a staggered schedule
+            if (n >= warm_min and st is not None
+                    and st[1] % reanchor != 0):
+                groups.setdefault(n, []).append((i, key))
+        for n, items in groups.items():
+            A = _torch.stack([mats[i] for i, _ in items])
+            V0 = _torch.stack([self._warm_state[k][0] for _, k in items])
+            d, V, off, _iters = block_jacobi_eigh_batched(
+                A, V0=V0, tol=tol, max_iters=max_iters)
+            offh = off.cpu()
+            for j, (i, k) in enumerate(items):
+                if float(offh[j]) < tol:
+                    results[i] = (d[j], V[j])
+                    self._warm_state[k][0] = V[j].contiguous()
+                    self._warm_state[k][1] += 1
+                else:
+                    # basis genuinely rotated: cold-solve below
+                    self._warm_state.pop(k, None)
+        rest = [i for i in range(len(mats)) if results[i] is None]
+        if rest:
+            lib = mat_eig_multi([mats[i] for i in rest],
+                                need_sorted=False)
+            for i, r in zip(rest, lib):
+                results[i] = r
+                n = int(mats[i].shape[-1])
+                if n >= warm_min:
+                    key = (id(work[i][0]), work[i][1])
+                    st = self._warm_state.get(key)
+                    # stagger each key's re-anchor phase so no single
+                    # step pays every exact solve at once
+                    age = st[1] + 1 if st else \
+                        1 + (len(self._warm_state) * 7) % reanchor
+                    self._warm_state[key] = [r[1].contiguous(), age]
+        return results
 
 
 class KFACEigen(EigenComputeMixin, KFACInverse):
