@@ -114,14 +114,23 @@ class EigenComputeMixin:
             d, V, off, _iters = block_jacobi_eigh_batched(
                 A, V0=V0, tol=tol, max_iters=max_iters)
             offh = off.cpu()
+            if hasattr(self, "phase_times"):  # KFAC_PHASE_TIMING stats
+                self.phase_times["warm_eig_iters"] = \
+                    self.phase_times.get("warm_eig_iters", 0) + _iters
             for j, (i, k) in enumerate(items):
                 if float(offh[j]) < tol:
                     results[i] = (d[j], V[j])
                     self._warm_state[k][0] = V[j].contiguous()
                     self._warm_state[k][1] += 1
+                    if hasattr(self, "phase_times"):
+                        self.phase_times["warm_eig_hits"] = \
+                            self.phase_times.get("warm_eig_hits", 0) + 1
                 else:
                     # basis genuinely rotated: cold-solve below
                     self._warm_state.pop(k, None)
+                    if hasattr(self, "phase_times"):
+                        self.phase_times["warm_eig_miss"] = \
+                            self.phase_times.get("warm_eig_miss", 0) + 1
         rest = [i for i in range(len(mats)) if results[i] is None]
         if rest:
             lib = mat_eig_multi([mats[i] for i in rest],

@@ -220,6 +220,7 @@ def test_custom_sytrd_precondition_accuracy(solver, monkeypatch):
         pytest.skip("old _kfac_solver build")
     from kfac_pytorch_amd.ops.linalg import mat_eig_multi
     monkeypatch.setenv("KFAC_SYTRD_MIN", "1000")
+    monkeypatch.setenv("KFAC_CUSTOM_SYTRD", "1")
     a = spd(1536, seed=5)
     (res,) = [mat_eig_multi([a], need_sorted=False)[0]]
     torch.cuda.synchronize()
