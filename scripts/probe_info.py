@@ -8,7 +8,6 @@ os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 os.environ.setdefault("MASTER_PORT", "29791")
 if not dist.is_initialized():
     dist.init_process_group("gloo", world_size=1, rank=0, init_method="env://")
-import torchvision
 import kfac_pytorch_amd as kfac
 from kfac_pytorch_amd.ops import linalg
 
@@ -19,11 +18,8 @@ def spy(info):
     orig(info)
 linalg._defer_info = spy
 
-try:
-    model = torchvision.models.resnet50().cuda()
-except Exception:
-    from kfac_pytorch_amd.models.imagenet_resnet import resnet50
-    model = resnet50().cuda()
+from kfac_pytorch_amd.models.imagenet_resnet import resnet50
+model = resnet50().cuda()
 pre = kfac.KFAC_EIGEN_DP(model, damping=0.002)
 data = torch.randn(32, 3, 224, 224, device="cuda")
 tgt = torch.randint(0, 1000, (32,), device="cuda")
