@@ -9,6 +9,8 @@ os.environ.setdefault("MASTER_PORT", "29791")
 if not dist.is_initialized():
     dist.init_process_group("gloo", world_size=1, rank=0, init_method="env://")
 import kfac_pytorch_amd as kfac
+from kfac_pytorch_amd.parallel import comm as kcomm
+kcomm.init()
 from kfac_pytorch_amd.ops import linalg
 
 calls = []
