@@ -266,6 +266,10 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
         device = mats[rest[0]].device
         sytrd_min = int(os.environ.get("KFAC_SYTRD_MIN",
                                        SYTRD_DISPATCH_MIN))
+        # info words of slot-issued calls are written on POOL streams:
+        # accumulate them only after the join orders the torch stream
+        # behind the pool, else the check reads uninitialized memory
+        pending_infos = []
         use_custom = _custom_sytrd_on()
         custom_jobs = []  # (members, n4, stacked, tauT, D, Cs)
         for n, members in _pad_buckets(dims):
@@ -292,7 +296,7 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                     except RuntimeError:
                         singles.extend(i for _, i in members)
                         continue
-                    _defer_info(_info)
+                    pending_infos.append(_info)
                     slot += 1
                     issued_on_pool = True
                     for k, (m, i) in enumerate(members):
@@ -301,12 +305,12 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                         out[i] = (W[k, pad:],
                                   Q[pad:, pad:] if pad else Q)
                     continue
-                _defer_info(status)
+                pending_infos.append(status)
                 D = stacked.diagonal(dim1=1, dim2=2).contiguous()
                 Cs = []
                 for k in range(b):
                     C, cinfo = solver.stedc_slot_(D[k], E[k], slot % 8)
-                    _defer_info(cinfo)
+                    pending_infos.append(cinfo)
                     slot += 1
                     Cs.append(C)
                 issued_on_pool = True
@@ -331,7 +335,7 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
             except RuntimeError:
                 singles.extend(i for _, i in members)
                 continue
-            _defer_info(_info)
+            pending_infos.append(_info)
             slot += 1
             issued_on_pool = True
             for k, (m, i) in enumerate(members):
@@ -361,6 +365,8 @@ def mat_eig_multi(mats, method: str = "auto", need_sorted: bool = True):
                     out[i] = mat_eig(mats[i], method="eigh")
         elif issued_on_pool:
             solver.join_pool_()
+        for _pi in pending_infos:
+            _defer_info(_pi)
         # WY back-transform of the custom-sytrd buckets (torch stream,
         # after the pool join so every stedc has finished)
         for members, n4, stacked, tauT, D, Cs in custom_jobs:
@@ -429,6 +435,7 @@ def mat_inv_multi(mats, damp_diag=None):
         slot = 0
         issued = False
         pending = []
+        pending_infos = []
         for n, members in _pad_buckets(dims):
             if len(members) < 2:
                 singles.extend(i for _, i in members)
@@ -445,7 +452,7 @@ def mat_inv_multi(mats, damp_diag=None):
             except RuntimeError:
                 singles.extend(i for _, i in members)
                 continue
-            _defer_info(_info)
+            pending_infos.append(_info)
             slot += 1
             issued = True
             pending.append((stacked, members))
@@ -465,6 +472,8 @@ def mat_inv_multi(mats, damp_diag=None):
                 singles = []
         elif issued:
             solver.join_pool_()
+        for _pi in pending_infos:
+            _defer_info(_pi)
         # mirror AFTER the join so the torch stream sees finished potri
         for stacked, members in pending:
             n = stacked.shape[-1]
