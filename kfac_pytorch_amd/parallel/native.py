@@ -119,3 +119,26 @@ class NativeCommunicator:
         """Stream-ordered drain: the torch current stream waits on every
         comm stream via HIP events; the host does not block."""
         self._impl.join()
+
+
+def fused_eigen_multibcast(nat: "NativeCommunicator", factors,
+                           packed_out) -> None:
+    """The reference's tcmm research path, natively: one fused
+    compute-then-broadcast pass over a list of symmetric factors
+    (reference: packages/tcmm/src/communicator.cpp:75-117 driven by
+    scripts/bench_ops.py:111-146).  Each factor >= 512^2 elements is
+    eigendecomposed by its round-robin owner (the per-factor callback)
+    and the packed ``(n, n+1)`` output ``[Q | d]`` is broadcast from
+    that owner on a rotating RCCL comm; smaller factors are computed
+    redundantly on every rank (a ~64^2 broadcast on xGMI costs more
+    than the redundant solve).  Call ``nat.join()`` is included."""
+    from kfac_pytorch_amd.ops.linalg import mat_eig
+
+    def op(inp, out):
+        d, Q = mat_eig(inp, method="auto")
+        n = inp.shape[0]
+        out[:, :n].copy_(Q)
+        out[:, n] = d
+
+    nat.multi_bcast(list(factors), list(packed_out), op)
+    nat.join()

@@ -416,6 +416,15 @@ class KFACBase(optim.Optimizer):
         if not self._state_ready:
             self._init_state()
             self._state_ready = True
+            # create the rotating duplicate communicators EAGERLY and
+            # collectively here (every rank passes this point at step 0
+            # with identical state) instead of lazily inside the first
+            # broadcast -- group creation is itself a collective, and a
+            # lazy first-use site is a hang hazard if any rank's call
+            # pattern ever diverged
+            if (self.comm.size() > 1 and self._native_comm() is None
+                    and not self.exclude_communicate_inverse):
+                self.comm.ensure_rotating_groups()
 
         if self.steps % self.fac_update_freq == 0:
             if not self.exclude_compute_factor:

@@ -201,13 +201,56 @@ class KFACEigen(EigenComputeMixin, KFACInverse):
             self.m_dG[m] = self._owner_view(self.eig_buckets, f"dG{i}")
 
     # ------------------------------------------------------------- inverses
+    def _use_multibcast(self) -> bool:
+        """KFAC_NATIVE_MULTIBCAST=1: the reference's fused tcmm
+        research path (per-factor eigensolve on the multi_bcast owner,
+        overlapped rotating-comm broadcasts of packed outputs) instead
+        of the default batched-solve + flat-bucket broadcast.  Needs
+        the native communicator (world > 1, GPU, KFAC_NATIVE_COMM=1)."""
+        import os
+        return (os.environ.get("KFAC_NATIVE_MULTIBCAST") == "1"
+                and self._native_comm() is not None)
+
     def _compute_inverse(self):
         """Owner ranks eigendecompose their factors (batched); eigenvalues
         clamped at eps (reference :98-119)."""
+        if self._use_multibcast():
+            return  # fused into _communicate_inverse
         self._eigendecompose_owned()
 
     def _communicate_inverse(self):
+        if self._use_multibcast():
+            self._fused_eigen_multibcast()
+            return
         self._broadcast_owner_buckets(self.eig_buckets)
+
+    def _fused_eigen_multibcast(self):
+        """Fused compute+broadcast of every factor's eigendecomposition
+        (reference research path: communicator.cpp:75-117,
+        scripts/bench_ops.py:111-146).  The multi_bcast round-robin
+        owner schedule replaces module_ranks for THIS phase; outputs
+        land identically on every rank, which is all the MPD eigen
+        pred phase needs."""
+        from kfac_pytorch_amd.parallel.native import \
+            fused_eigen_multibcast
+        nat = self._native_comm()
+        facs, outs, metas = [], [], []
+        for m in self.modules:
+            for kind, fac in (("A", self.m_A[m]), ("G", self.m_G[m])):
+                n = fac.shape[0]
+                facs.append(fac)
+                outs.append(fac.new_empty(n, n + 1))
+                metas.append((m, kind, n))
+        fused_eigen_multibcast(nat, facs, outs)
+        for (m, kind, n), out in zip(metas, outs):
+            d = out[:, n]
+            clamped = d * (d > self.eps)
+            if kind == "A":
+                self.m_QA[m].copy_(out[:, :n])
+                self.m_dA[m].copy_(clamped)
+            else:
+                self.m_QG[m].copy_(out[:, :n])
+                self.m_dG[m].copy_(clamped)
 
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):

@@ -418,3 +418,111 @@ def _worker_exclude_comm_parts(rank, world, tmpfile):
 
 def test_exclude_comm_parts_no_deadlock():
     _run_spawn(_worker_exclude_comm_parts)
+
+
+# --------------------------------------------------------------------------
+# world-8 coverage: every algorithm at the driver's SCALE width, with
+# factor-wise distribution engaged (world > #modules) and both
+# communicate_inverse modes -- no multi-rank codepath left CPU-untested
+# at the scale the 8-GPU run uses (gloo here, RCCL there).
+# --------------------------------------------------------------------------
+def _worker_world8_algorithms(rank, world, tmpfile, name):
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    torch.manual_seed(21)
+    model = MLP()
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+    kwargs = dict(damping=0.01)
+    pre = kfac.get_kfac_module(name)(model, **kwargs)
+    x, y = _global_batch(seed=rank + 70)
+    for step in range(2):
+        _train_grads(model, x, y)
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        pre.step()
+        for p in model.parameters():
+            mine = p.grad.clone()
+            comm.broadcast(p.grad.data, src=0)
+            torch.testing.assert_close(mine, p.grad,
+                                       rtol=1e-4, atol=1e-5)
+    if name == "eigen":
+        # world 8 > 3 modules -> factor-wise: rank_g = rank_a + 1
+        ras = sorted(ra for ra, _ in pre.module_ranks.values())
+        rgs = [rg for _, rg in pre.module_ranks.values()]
+        assert any(ra != rg for (ra, rg) in pre.module_ranks.values())
+        assert len(set(ras)) == len(ras)  # distinct owners at world 8
+        del rgs
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("name", ["eigen", "eigen_dp", "inverse",
+                                  "inverse_dp"])
+def test_world8_all_algorithms(name):
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(_worker_world8_algorithms, args=(8, tmpfile, name),
+             nprocs=8, join=True)
+
+
+def _worker_world8_inverse_modes(rank, world, tmpfile):
+    """'inverse' with communicate_inverse_or_not both ways at world 8:
+    the two modes must produce identical preconditioned gradients."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    grads = {}
+    for mode in (False, True):
+        torch.manual_seed(33)
+        model = MLP()
+        for p in model.parameters():
+            comm.broadcast(p.data, src=0)
+        pre = kfac.KFAC_INV(model, damping=0.01,
+                            communicate_inverse_or_not=mode)
+        x, y = _global_batch(seed=rank + 90)
+        _train_grads(model, x, y)
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        pre.step()
+        grads[mode] = [p.grad.clone() for p in model.parameters()]
+    for a, b in zip(grads[False], grads[True]):
+        torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
+    dist.destroy_process_group()
+
+
+def test_world8_inverse_both_comm_modes():
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(_worker_world8_inverse_modes, args=(8, tmpfile),
+             nprocs=8, join=True)
+
+
+def _worker_rotating_fewer_than_world(rank, world, tmpfile):
+    """Rotating duplicate-group count != world size: broadcasts rooted
+    at every rank must still land (groups are WORLD duplicates, the
+    rotation only spreads concurrent traffic)."""
+    comm = _init_worker(rank, world, tmpfile)
+    n = comm.ensure_rotating_groups(3)
+    assert n == 3
+    handles = []
+    tensors = []
+    for r in range(world):
+        t = torch.full((16,), float(rank), dtype=torch.float32)
+        if rank == r:
+            t.fill_(100.0 + r)
+        tensors.append(t)
+        handles.append(comm.broadcast_async_(
+            t, src=r, group=comm.rotating_group(r)))
+    comm.synchronize(handles)
+    for r, t in enumerate(tensors):
+        torch.testing.assert_close(t, torch.full((16,), 100.0 + r))
+    dist.destroy_process_group()
+
+
+def test_rotating_groups_fewer_than_world():
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(_worker_rotating_fewer_than_world, args=(4, tmpfile),
+             nprocs=4, join=True)
