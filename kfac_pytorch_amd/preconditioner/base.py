@@ -144,11 +144,63 @@ class KFACBase(optim.Optimizer):
         """Save the module input (a) by reference."""
         if self._save_input_enabled():
             self.m_a[module] = input[0].data
+            self._overlap_factor(module, "A", input[0].data)
 
     def _backward_hook_event(self, module, grad_input, grad_output):
         """Save the grad wrt output (g) by reference."""
         if self._save_grad_enabled():
             self.m_g[module] = grad_output[0].data
+            self._overlap_factor(module, "G", grad_output[0].data)
+
+    # ------------------------------------------ factor/backward overlap
+    def _overlap_enabled(self) -> bool:
+        """KFAC_FACTOR_OVERLAP (default on, GPU only): compute each
+        layer's FRESH factor on a side stream the moment its hook
+        fires, so the batch-sized factor products (im2col + MFMA SYRK)
+        hide under the rest of forward/backward instead of running as
+        a serial phase after it.  ``step()`` then only applies the
+        running-average AXPY.  Numerics identical to the inline path
+        (fresh factor staged, last capture wins on hook refires)."""
+        if not hasattr(self, "_ov_flag"):
+            import os
+            self._ov_flag = (os.environ.get("KFAC_FACTOR_OVERLAP", "1")
+                             != "0" and torch.cuda.is_available())
+        return self._ov_flag
+
+    def _overlap_factor(self, module, kind, tensor):
+        if not self._overlap_enabled() or not tensor.is_cuda:
+            return
+        if not hasattr(self, "_ov_stream"):
+            self._ov_stream = torch.cuda.Stream()
+            self._ov_ready = torch.cuda.Event()
+            self._ov_fresh: Dict = {}   # (module, kind) -> staging
+            self._ov_new = set()        # keys computed since last consume
+        key = (module, kind)
+        self._ov_ready.record()
+        self._ov_stream.wait_event(self._ov_ready)
+        with torch.cuda.stream(self._ov_stream):
+            # the capture is allocated on the default stream; pin it
+            # until the side-stream factor product finishes
+            tensor.record_stream(self._ov_stream)
+            stg = self._ov_fresh.get(key)
+            if kind == "A":
+                res = self.computeA(tensor, module, out=stg, decay=None)
+            else:
+                res = self.computeG(tensor, module, batch_averaged=True,
+                                    out=stg, decay=None)
+            self._ov_fresh[key] = res
+        self._ov_new.add(key)
+
+    def _consume_overlapped(self):
+        """Keys side-computed since the last call; the default stream
+        is ordered behind the side stream before returning."""
+        new = getattr(self, "_ov_new", None)
+        if not new:
+            return frozenset()
+        torch.cuda.current_stream().wait_stream(self._ov_stream)
+        out = frozenset(new)
+        new.clear()
+        return out
 
     def _register_module_hooks(self, model: nn.Module):
         name_idx = 0

@@ -95,12 +95,24 @@ class KFACInverse(KFACBase):
     def _compute_factors(self):
         """Local factors + running average, every rank, every layer
         (reference :80-91); the running-average update is fused into the
-        factor kernel via out=/decay=."""
+        factor kernel via out=/decay= -- except for layers whose fresh
+        factor was already side-computed under backward
+        (_overlap_factor), which only need the AXPY here."""
+        from kfac_pytorch_amd.ops.factors import update_running_avg
+        done = self._consume_overlapped()
         for m in self.modules:
-            self.computeA(self.m_a[m], m, out=self.m_A[m],
-                          decay=self.factor_decay)
-            self.computeG(self.m_g[m], m, batch_averaged=True,
-                          out=self.m_G[m], decay=self.factor_decay)
+            if (m, "A") in done:
+                update_running_avg(self._ov_fresh[(m, "A")],
+                                   self.m_A[m], self.factor_decay)
+            else:
+                self.computeA(self.m_a[m], m, out=self.m_A[m],
+                              decay=self.factor_decay)
+            if (m, "G") in done:
+                update_running_avg(self._ov_fresh[(m, "G")],
+                                   self.m_G[m], self.factor_decay)
+            else:
+                self.computeG(self.m_g[m], m, batch_averaged=True,
+                              out=self.m_G[m], decay=self.factor_decay)
 
     def _communicate_factors(self):
         """ONE flat allreduce-average for all layers' A and G

@@ -47,6 +47,7 @@ class KFACInverseDP(KFACInverse):
             rank_a, _ = self.module_ranks[module]
             if self.comm.rank() == rank_a:
                 self.m_a[module] = input[0].data
+                self._overlap_factor(module, "A", input[0].data)
 
     def _backward_hook_event(self, module, grad_input, grad_output):
         """Save grad-output only on the owner rank (reference :67-72)."""
@@ -54,6 +55,7 @@ class KFACInverseDP(KFACInverse):
             _, rank_g = self.module_ranks[module]
             if self.comm.rank() == rank_g:
                 self.m_g[module] = grad_output[0].data
+                self._overlap_factor(module, "G", grad_output[0].data)
 
     # ---------------------------------------------------------------- state
     def _init_state(self):
@@ -62,16 +64,28 @@ class KFACInverseDP(KFACInverse):
 
     # -------------------------------------------------------------- factors
     def _compute_factors(self):
-        """Owner-local factors only (reference :75-90)."""
+        """Owner-local factors only (reference :75-90); layers whose
+        fresh factor was already side-computed under backward
+        (_overlap_factor) only get the running-average AXPY here."""
+        from kfac_pytorch_amd.ops.factors import update_running_avg
+        done = self._consume_overlapped()
         rank = self.comm.rank()
         for m in self.modules:
             rank_a, rank_g = self.module_ranks[m]
             if rank == rank_a:
-                self.computeA(self.m_a[m], m, out=self.m_A[m],
-                              decay=self.factor_decay)
+                if (m, "A") in done:
+                    update_running_avg(self._ov_fresh[(m, "A")],
+                                       self.m_A[m], self.factor_decay)
+                else:
+                    self.computeA(self.m_a[m], m, out=self.m_A[m],
+                                  decay=self.factor_decay)
             if rank == rank_g:
-                self.computeG(self.m_g[m], m, batch_averaged=True,
-                              out=self.m_G[m], decay=self.factor_decay)
+                if (m, "G") in done:
+                    update_running_avg(self._ov_fresh[(m, "G")],
+                                       self.m_G[m], self.factor_decay)
+                else:
+                    self.computeG(self.m_g[m], m, batch_averaged=True,
+                                  out=self.m_G[m], decay=self.factor_decay)
 
     def _communicate_factors(self):
         """No factor communication -- the whole point of DP-KFAC

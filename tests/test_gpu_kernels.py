@@ -258,3 +258,50 @@ def test_eigen_precondition_multi_matches_single():
         ref = eigen_precondition(QAs[i], dAs[i], QGs[i], dGs[i],
                                  grads[i].clone(), 0.002)
         torch.testing.assert_close(outs[i], ref, rtol=1e-4, atol=1e-5)
+
+
+def test_factor_overlap_matches_inline():
+    """KFAC_FACTOR_OVERLAP side-stream factor computation must produce
+    bitwise-compatible factors and preconditioned grads vs the inline
+    phase (same kernels, different stream/timing)."""
+    import os
+    import torch.nn as nn
+    import torch.nn.functional as F
+    import kfac_pytorch_amd as kfac
+    import kfac_pytorch_amd.parallel.comm as comm_mod
+    import torch.distributed as dist
+    from tests.conftest import free_port
+    if not dist.is_initialized():
+        dist.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{free_port()}",
+            world_size=1, rank=0)
+    comm_mod.reset()
+    comm_mod.init("Torch")
+
+    def run(flag):
+        os.environ["KFAC_FACTOR_OVERLAP"] = flag
+        torch.manual_seed(3)
+        model = nn.Sequential(
+            nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(), nn.Flatten(),
+            nn.Linear(8 * 8 * 8, 10)).cuda()
+        pre = kfac.KFAC_EIGEN_DP(model, damping=0.01)
+        x = torch.randn(4, 3, 8, 8, device="cuda")
+        y = torch.randint(0, 10, (4,), device="cuda")
+        for _ in range(3):
+            model.zero_grad(set_to_none=False)
+            F.cross_entropy(model(x), y).backward()
+            pre.step()
+        torch.cuda.synchronize()
+        A = {i: pre.m_A[m].clone() for i, m in enumerate(pre.modules)}
+        G = {i: pre.m_G[m].clone() for i, m in enumerate(pre.modules)}
+        grads = [p.grad.clone() for p in model.parameters()]
+        os.environ.pop("KFAC_FACTOR_OVERLAP", None)
+        return A, G, grads
+
+    A1, G1, g1 = run("1")
+    A0, G0, g0 = run("0")
+    for k in A1:
+        torch.testing.assert_close(A1[k], A0[k], rtol=1e-5, atol=1e-6)
+        torch.testing.assert_close(G1[k], G0[k], rtol=1e-5, atol=1e-6)
+    for a, b in zip(g1, g0):
+        torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
