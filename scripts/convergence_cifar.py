@@ -45,13 +45,28 @@ def run(opt_name, args, xtr, ytr, xte, yte, device):
 
     torch.manual_seed(7)
     model = get_cifar_model(args.model).to(device)
-    optimizer = torch.optim.SGD(model.parameters(), lr=0.1,
+    base_lr = 0.05
+    optimizer = torch.optim.SGD(model.parameters(), lr=base_lr,
                                 momentum=0.9, weight_decay=5e-4)
     precond = None
     if opt_name == "kfac":
-        precond = kfac.KFAC_EIGEN_DP(model, lr=0.1, damping=0.03,
+        precond = kfac.KFAC_EIGEN_DP(model, lr=base_lr, damping=0.03,
                                      fac_update_freq=1,
                                      kfac_update_freq=10)
+    # warmup + step decay at 50%/75% (the reference's CIFAR recipe
+    # shape, examples/pytorch_cifar10_resnet.py lr schedule)
+    def lr_fn(epoch):
+        if epoch < 1:
+            return 0.2 + 0.8 * epoch
+        f = 1.0
+        if epoch >= args.epochs // 2:
+            f *= 0.1
+        if epoch >= (3 * args.epochs) // 4:
+            f *= 0.1
+        return f
+    scheds = [torch.optim.lr_scheduler.LambdaLR(optimizer, lr_fn)]
+    if precond is not None:
+        scheds.append(torch.optim.lr_scheduler.LambdaLR(precond, lr_fn))
     n = xtr.shape[0]
     bs = args.batch_size
     hist = []
@@ -70,6 +85,8 @@ def run(opt_name, args, xtr, ytr, xte, yte, device):
             if precond is not None:
                 precond.step()
             optimizer.step()
+        for s in scheds:
+            s.step()
         if device.type == "cuda":
             torch.cuda.synchronize()
         wall = time.perf_counter() - t0
