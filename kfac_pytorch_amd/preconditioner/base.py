@@ -202,6 +202,45 @@ class KFACBase(optim.Optimizer):
         new.clear()
         return out
 
+    # --------------------------------------------- hipGraph phase capture
+    def _run_graphed(self, key, fn, fingerprint):
+        """Run ``fn`` (a launch-bound phase writing only into static
+        buffers) through a captured hipGraph, replaying on subsequent
+        steps.  ~300 small launches collapse into one graph launch.
+
+        ``fingerprint`` must cover everything baked into the capture:
+        scalar hyperparameters (damping) and the data_ptrs of every
+        input tensor whose STORAGE could be swapped between steps
+        (p.grad under zero_grad(set_to_none=True)) -- any change
+        triggers a clean recapture.  KFAC_PRED_GRAPH=0 disables."""
+        import os
+        if (not torch.cuda.is_available()
+                or os.environ.get("KFAC_PRED_GRAPH", "1") == "0"
+                or getattr(self, "_graph_disabled", False)):
+            fn()
+            return
+        if not hasattr(self, "_graphs"):
+            self._graphs: Dict = {}
+        entry = self._graphs.get(key)
+        if entry is not None and entry[1] == fingerprint:
+            entry[0].replay()
+            return
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                fn()  # warmup (allocator state, lazy inits)
+            torch.cuda.current_stream().wait_stream(side)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                fn()
+            self._graphs[key] = (graph, fingerprint)
+            graph.replay()
+        except Exception:
+            # capture-unfriendly op somewhere: stay eager for good
+            self._graph_disabled = True
+            fn()
+
     def _register_module_hooks(self, model: nn.Module):
         name_idx = 0
         for module in model.modules():

@@ -255,12 +255,32 @@ class KFACEigen(EigenComputeMixin, KFACInverse):
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):
         """Implicit-eigen preconditioning on every rank (reference
-        :137-144), same-shape layers batched."""
-        preds = eigen_precondition_multi(
-            [self.m_QA[m] for m in self.modules],
-            [self.m_dA[m] for m in self.modules],
-            [self.m_QG[m] for m in self.modules],
-            [self.m_dG[m] for m in self.modules],
-            [self._get_grad(m) for m in self.modules], self.damping)
-        for m, p in zip(self.modules, preds):
-            self.m_precon_grad[m] = p
+        :137-144), same-shape layers batched, replayed as one hipGraph
+        (launch-bound phase; see base._run_graphed)."""
+        mods = self.modules
+
+        def fn():
+            preds = eigen_precondition_multi(
+                [self.m_QA[m] for m in mods],
+                [self.m_dA[m] for m in mods],
+                [self.m_QG[m] for m in mods],
+                [self.m_dG[m] for m in mods],
+                [self._get_grad(m) for m in mods], self.damping)
+            for m, p in zip(mods, preds):
+                if m in self.m_precon_grad and \
+                        self.m_precon_grad[m].shape == p.shape:
+                    self.m_precon_grad[m].copy_(p)
+                else:
+                    self.m_precon_grad[m] = p
+
+        if any(m not in self.m_precon_grad for m in mods):
+            # first step allocates the output buffers eagerly so the
+            # captured graph only ever writes into static storage
+            fn()
+            return
+        fp = (float(self.damping),
+              tuple(m.weight.grad.data_ptr() for m in mods),
+              tuple(m.bias.grad.data_ptr() for m in mods
+                    if m.bias is not None),
+              tuple(self.m_QA[m].data_ptr() for m in mods))
+        self._run_graphed("pred", fn, fp)

@@ -85,9 +85,23 @@ class KFACEigenDP(EigenComputeMixin, KFACInverseDP):
                 owned.append(m)
         if not owned:
             return
-        preds = eigen_precondition_multi(
-            [self.m_QA[m] for m in owned], [self.m_dA[m] for m in owned],
-            [self.m_QG[m] for m in owned], [self.m_dG[m] for m in owned],
-            [self._get_grad(m) for m in owned], self.damping)
-        for m, p in zip(owned, preds):
-            self.m_precon_grad[m].copy_(p)
+
+        def fn():
+            preds = eigen_precondition_multi(
+                [self.m_QA[m] for m in owned],
+                [self.m_dA[m] for m in owned],
+                [self.m_QG[m] for m in owned],
+                [self.m_dG[m] for m in owned],
+                [self._get_grad(m) for m in owned], self.damping)
+            for m, p in zip(owned, preds):
+                self.m_precon_grad[m].copy_(p)
+
+        # the ~300 small launches of the grouped-bmm pred phase replay
+        # as one hipGraph; recaptured when damping or any grad storage
+        # changes (see base._run_graphed)
+        fp = (float(self.damping),
+              tuple(m.weight.grad.data_ptr() for m in owned),
+              tuple(m.bias.grad.data_ptr() for m in owned
+                    if m.bias is not None),
+              tuple(self.m_QA[m].data_ptr() for m in owned))
+        self._run_graphed("pred", fn, fp)

@@ -305,3 +305,47 @@ def test_factor_overlap_matches_inline():
         torch.testing.assert_close(G1[k], G0[k], rtol=1e-5, atol=1e-6)
     for a, b in zip(g1, g0):
         torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
+
+
+def test_pred_graph_matches_eager():
+    """hipGraph-captured pred phase must match the eager phase across
+    steps, damping changes (recapture) and both eigen families."""
+    import os
+    import torch.nn as nn
+    import torch.nn.functional as F
+    import kfac_pytorch_amd as kfac
+    import kfac_pytorch_amd.parallel.comm as comm_mod
+    import torch.distributed as dist
+    from tests.conftest import free_port
+    if not dist.is_initialized():
+        dist.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{free_port()}",
+            world_size=1, rank=0)
+    comm_mod.reset()
+    comm_mod.init("Torch")
+
+    def run(flag, name):
+        os.environ["KFAC_PRED_GRAPH"] = flag
+        torch.manual_seed(5)
+        model = nn.Sequential(
+            nn.Conv2d(3, 8, 3, padding=1), nn.ReLU(), nn.Flatten(),
+            nn.Linear(8 * 8 * 8, 10)).cuda()
+        pre = kfac.get_kfac_module(name)(model, damping=0.01)
+        x = torch.randn(4, 3, 8, 8, device="cuda")
+        y = torch.randint(0, 10, (4,), device="cuda")
+        for step in range(4):
+            if step == 2:  # force a recapture mid-run
+                pre.param_groups[0]["damping"] = 0.02
+            model.zero_grad(set_to_none=False)
+            F.cross_entropy(model(x), y).backward()
+            pre.step()
+        torch.cuda.synchronize()
+        out = [p.grad.clone() for p in model.parameters()]
+        os.environ.pop("KFAC_PRED_GRAPH", None)
+        return out
+
+    for name in ("eigen_dp", "eigen"):
+        g1 = run("1", name)
+        g0 = run("0", name)
+        for a, b in zip(g1, g0):
+            torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
