@@ -526,3 +526,55 @@ def test_rotating_groups_fewer_than_world():
     os.unlink(tmpfile)
     mp.spawn(_worker_rotating_fewer_than_world, args=(4, tmpfile),
              nprocs=4, join=True)
+
+
+def _worker_lpt_balance(rank, world, tmpfile):
+    """Cost-aware (LPT) default schedule: rank eigensolve loads must be
+    far more even than round-robin's on a skewed layer list, and every
+    rank must compute the identical assignment."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+
+    class Skewed(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.big = nn.Linear(512, 8)     # dominates (512^3)
+            self.s1 = nn.Linear(8, 8)
+            self.s2 = nn.Linear(8, 8)
+            self.s3 = nn.Linear(8, 8)
+
+        def forward(self, x):
+            return self.s3(self.s2(self.s1(self.big(x))))
+
+    torch.manual_seed(0)
+    model = Skewed()
+    pre = kfac.KFAC_EIGEN_DP(model, damping=0.01)
+    from kfac_pytorch_amd.ops.factors import factor_dims
+    loads = [0] * world
+    for m, (ra, rg) in pre.module_ranks.items():
+        assert ra == rg
+        da, dg = factor_dims(m)
+        loads[ra] += da ** 3 + dg ** 3
+    # the big layer must be alone on its rank (3 small layers elsewhere)
+    big_rank = pre.module_ranks[model.big][0]
+    assert sum(1 for m, (ra, _) in pre.module_ranks.items()
+               if ra == big_rank) == 1, pre.module_ranks
+    # identical schedule on every rank
+    mine = torch.tensor([pre.module_ranks[m][0] for m in pre.modules],
+                        dtype=torch.float32)
+    ref = mine.clone()
+    comm.broadcast(ref, src=0)
+    torch.testing.assert_close(mine, ref)
+    # and a step must still work end-to-end
+    x = torch.randn(8, 512)
+    y = torch.randint(0, 8, (8,))
+    model.zero_grad(set_to_none=False)
+    F.cross_entropy(model(x), y).backward()
+    for p in model.parameters():
+        comm.allreduce(p.grad.data, op=comm.Average)
+    pre.step()
+    dist.destroy_process_group()
+
+
+def test_lpt_schedule_balances_and_agrees():
+    _run_spawn(_worker_lpt_balance)
