@@ -514,8 +514,29 @@ def eigen_precondition(QA: torch.Tensor, dA: torch.Tensor,
 def inverse_precondition(inv_A: torch.Tensor, inv_G: torch.Tensor,
                          grad: torch.Tensor) -> torch.Tensor:
     """P = inv_G @ grad @ inv_A
-    (reference: kfac/kfac_preconditioner_inv.py:156-161)."""
+    (reference: kfac/kfac_preconditioner_inv.py:156-161).
+    Accepts the group-stacked 3-D form (batched bmm per group block)."""
+    if grad.dim() == 3:
+        return torch.bmm(inv_G, torch.bmm(grad, inv_A))
     return inv_G @ grad @ inv_A
+
+
+def eigen_precondition_grouped(QA: torch.Tensor, dA: torch.Tensor,
+                               QG: torch.Tensor, dG: torch.Tensor,
+                               grad: torch.Tensor,
+                               damping: float) -> torch.Tensor:
+    """Implicit-eigen preconditioning of a grouped conv's
+    group-stacked gradient: QA/dA/QG/dG are (g, ., .) per-group
+    eigenbases, grad is (g, dG, dA); every op is one batched bmm plus
+    the fused batched denominator kernel on GPU."""
+    v1 = torch.bmm(torch.bmm(QG.mT, grad), QA).contiguous()
+    if grad.is_cuda:
+        from kfac_pytorch_amd.ops import _ext
+        _ext.eigen_scale_batched_(v1, dG.contiguous(), dA.contiguous(),
+                                  float(damping))
+    else:
+        v1 = v1 / (dG.unsqueeze(-1) * dA.unsqueeze(-2) + damping)
+    return torch.bmm(torch.bmm(QG, v1), QA.mT)
 
 
 def eigen_precondition_multi(QAs, dAs, QGs, dGs, grads, damping: float):

@@ -251,8 +251,9 @@ class KFACBase(optim.Optimizer):
                     and classname == 'Linear'
                     and module.out_features == self.exclude_vocabulary_size):
                 continue  # exclude pre-softmax vocab projection (ref :139-140)
-            if classname == 'Conv2d' and module.groups != 1:
-                continue  # grouped conv factors don't match weight layout
+            # grouped convs get exact block-diagonal factors (one
+            # block per group; beyond the reference, which computes a
+            # single WRONG dense factor for groups > 1)
             if (classname == 'Linear' and module.out_features >= 10000
                     and self.exclude_vocabulary_size is None):
                 logger.warning(
@@ -308,11 +309,12 @@ class KFACBase(optim.Optimizer):
                 module_ranks[module] = (rank_a, rank_g)
                 rank_iter += 1
         else:
-            from kfac_pytorch_amd.ops.factors import factor_dims
+            from kfac_pytorch_amd.ops.factors import (factor_dims,
+                                                      factor_groups)
             costs = []
             for i, m in enumerate(self.modules):
                 da, dg = factor_dims(m)
-                costs.append((da ** 3 + dg ** 3, i))
+                costs.append((factor_groups(m) * (da ** 3 + dg ** 3), i))
             # LPT: heaviest first into the currently lightest bin;
             # deterministic (stable sort + index tiebreak) so every
             # rank computes the identical schedule
@@ -356,18 +358,20 @@ class KFACBase(optim.Optimizer):
             rank_a, rank_g = self.module_ranks[m]
             da, dg = factor_dims(m)
             if not owner_only or rank == rank_a:
-                self.factor_bucket.add(f"A{i}", torch.Size((da, da)))
+                self.factor_bucket.add(
+                    f"A{i}", torch.Size(self._block_shape(m, da)))
             if not owner_only or rank == rank_g:
-                self.factor_bucket.add(f"G{i}", torch.Size((dg, dg)))
+                self.factor_bucket.add(
+                    f"G{i}", torch.Size(self._block_shape(m, dg)))
         self.factor_bucket.freeze(0.0)
         for i, m in enumerate(self.modules):
             if f"A{i}" in self.factor_bucket:
                 v = self.factor_bucket.view(f"A{i}")
-                v.diagonal().fill_(1.0)
+                v.diagonal(dim1=-2, dim2=-1).fill_(1.0)
                 self.m_A[m] = v
             if f"G{i}" in self.factor_bucket:
                 v = self.factor_bucket.view(f"G{i}")
-                v.diagonal().fill_(1.0)
+                v.diagonal(dim1=-2, dim2=-1).fill_(1.0)
                 self.m_G[m] = v
 
     def _alloc_owner_buckets(self, specs) -> List[FlatBucket]:
@@ -448,9 +452,28 @@ class KFACBase(optim.Optimizer):
         raise NotImplementedError
 
     # --------------------------------------------------------- grad plumbing
+    @staticmethod
+    def _block_shape(module, d: int, vec: bool = False):
+        """Factor-state shape: (d,)/(d,d), or group-stacked for grouped
+        convs (block-diagonal K-FAC)."""
+        from kfac_pytorch_amd.ops.factors import factor_groups
+        g = factor_groups(module)
+        if vec:
+            return (g, d) if g > 1 else (d,)
+        return (g, d, d) if g > 1 else (d, d)
+
     def _get_grad(self, module: nn.Module) -> torch.Tensor:
-        """Gradient as [out_dim, in_dim(+1)] (reference
+        """Gradient as [out_dim, in_dim(+1)] -- group-stacked
+        [g, out/g, in/g*kh*kw(+1)] for grouped convs (reference
         kfac/kfac_preconditioner_inv.py:145-154)."""
+        if isinstance(module, nn.Conv2d) and module.groups > 1:
+            g = module.groups
+            grad = module.weight.grad.data.view(
+                g, module.out_channels // g, -1)
+            if module.bias is not None:
+                grad = torch.cat(
+                    [grad, module.bias.grad.data.view(g, -1, 1)], 2)
+            return grad
         if isinstance(module, nn.Conv2d):
             grad = module.weight.grad.data.view(
                 module.weight.grad.data.size(0), -1)
@@ -461,8 +484,14 @@ class KFACBase(optim.Optimizer):
         return grad
 
     def _reshape_preconditioned_grad(self, module, v: torch.Tensor):
-        """Split [out, in(+1)] back into weight/bias shapes (reference
-        kfac/kfac_preconditioner_inv.py:178-186)."""
+        """Split [out, in(+1)] (or the group-stacked 3-D form) back into
+        weight/bias shapes (reference kfac_preconditioner_inv.py:178-186)."""
+        if v.dim() == 3:
+            if module.bias is not None:
+                vw = v[:, :, :-1].reshape(module.weight.grad.data.size())
+                vb = v[:, :, -1].reshape(module.bias.grad.data.size())
+                return [vw, vb]
+            return [v.reshape(module.weight.grad.data.size())]
         if module.bias is not None:
             vw = v[:, :-1].reshape(module.weight.grad.data.size())
             vb = v[:, -1:].reshape(module.bias.grad.data.size())

@@ -58,32 +58,40 @@ class EigenComputeMixin:
 
     def _eigendecompose_owned(self):
         import os
+        from kfac_pytorch_amd.ops.factors import factor_groups
         rank = self.comm.rank()
-        work = []
+        work = []  # (module, kind, group-index-or-None)
         for m in self.modules:
             rank_a, rank_g = self.module_ranks[m]
-            if rank == rank_a:
-                work.append((m, "A"))
-            if rank == rank_g:
-                work.append((m, "G"))
+            gr = factor_groups(m)
+            gis = range(gr) if gr > 1 else (None,)
+            for gi in gis:
+                if rank == rank_a:
+                    work.append((m, "A", gi))
+                if rank == rank_g:
+                    work.append((m, "G", gi))
         if not work:
             return
-        mats = [self.m_A[mod] if kind == "A" else self.m_G[mod]
-                for mod, kind in work]
+        mats = []
+        for mod, kind, gi in work:
+            t = self.m_A[mod] if kind == "A" else self.m_G[mod]
+            mats.append(t if gi is None else t[gi])
         if (os.environ.get("KFAC_WARM_EIG", "0") == "1"
                 and mats[0].is_cuda):
             results = self._warm_eig(work, mats)
         else:
             results = mat_eig_multi(mats, need_sorted=False)
 
-        for (mod, kind), (d, Q) in zip(work, results):
+        for (mod, kind, gi), (d, Q) in zip(work, results):
             clamped = d * (d > self.eps)
-            if kind == "A":
-                self.m_QA[mod].copy_(Q)
-                self.m_dA[mod].copy_(clamped)
+            tq = self.m_QA[mod] if kind == "A" else self.m_QG[mod]
+            td = self.m_dA[mod] if kind == "A" else self.m_dG[mod]
+            if gi is None:
+                tq.copy_(Q)
+                td.copy_(clamped)
             else:
-                self.m_QG[mod].copy_(Q)
-                self.m_dG[mod].copy_(clamped)
+                tq[gi].copy_(Q)
+                td[gi].copy_(clamped)
 
     def _warm_eig(self, work, mats):
         """Warm-tier dispatch: carried-basis blocked Jacobi for the big
@@ -100,8 +108,8 @@ class EigenComputeMixin:
             self._warm_state = {}  # (id(mod), kind) -> [V, age]
         results = [None] * len(mats)
         groups = {}
-        for i, (mod, kind) in enumerate(work):
-            key = (id(mod), kind)
+        for i, (mod, kind, gi) in enumerate(work):
+            key = (id(mod), kind, gi)
             st = self._warm_state.get(key)
             n = int(mats[i].shape[-1])
             # re-anchor exactly (library solve) on a staggered schedule
@@ -139,7 +147,7 @@ class EigenComputeMixin:
                 results[i] = r
                 n = int(mats[i].shape[-1])
                 if n >= warm_min:
-                    key = (id(work[i][0]), work[i][1])
+                    key = (id(work[i][0]), work[i][1], work[i][2])
                     st = self._warm_state.get(key)
                     # stagger each key's re-anchor phase so no single
                     # step pays every exact solve at once
@@ -189,10 +197,12 @@ class KFACEigen(EigenComputeMixin, KFACInverse):
         for i, m in enumerate(self.modules):
             rank_a, rank_g = self.module_ranks[m]
             da, dg = factor_dims(m)
-            specs.append((f"QA{i}", (da, da), rank_a))
-            specs.append((f"dA{i}", (da,), rank_a))
-            specs.append((f"QG{i}", (dg, dg), rank_g))
-            specs.append((f"dG{i}", (dg,), rank_g))
+            specs.append((f"QA{i}", self._block_shape(m, da), rank_a))
+            specs.append((f"dA{i}", self._block_shape(m, da, vec=True),
+                          rank_a))
+            specs.append((f"QG{i}", self._block_shape(m, dg), rank_g))
+            specs.append((f"dG{i}", self._block_shape(m, dg, vec=True),
+                          rank_g))
         self.eig_buckets = self._alloc_owner_buckets(specs)
         for i, m in enumerate(self.modules):
             self.m_QA[m] = self._owner_view(self.eig_buckets, f"QA{i}")
@@ -233,31 +243,40 @@ class KFACEigen(EigenComputeMixin, KFACInverse):
         pred phase needs."""
         from kfac_pytorch_amd.parallel.native import \
             fused_eigen_multibcast
+        from kfac_pytorch_amd.ops.factors import factor_groups
         nat = self._native_comm()
         facs, outs, metas = [], [], []
         for m in self.modules:
+            gr = factor_groups(m)
             for kind, fac in (("A", self.m_A[m]), ("G", self.m_G[m])):
-                n = fac.shape[0]
-                facs.append(fac)
-                outs.append(fac.new_empty(n, n + 1))
-                metas.append((m, kind, n))
+                for gi in (range(gr) if gr > 1 else (None,)):
+                    f2 = fac if gi is None else fac[gi]
+                    n = f2.shape[0]
+                    facs.append(f2.contiguous())
+                    outs.append(f2.new_empty(n, n + 1))
+                    metas.append((m, kind, gi, n))
         fused_eigen_multibcast(nat, facs, outs)
-        for (m, kind, n), out in zip(metas, outs):
+        for (m, kind, gi, n), out in zip(metas, outs):
             d = out[:, n]
             clamped = d * (d > self.eps)
-            if kind == "A":
-                self.m_QA[m].copy_(out[:, :n])
-                self.m_dA[m].copy_(clamped)
+            tq = self.m_QA[m] if kind == "A" else self.m_QG[m]
+            td = self.m_dA[m] if kind == "A" else self.m_dG[m]
+            if gi is None:
+                tq.copy_(out[:, :n])
+                td.copy_(clamped)
             else:
-                self.m_QG[m].copy_(out[:, :n])
-                self.m_dG[m].copy_(clamped)
+                tq[gi].copy_(out[:, :n])
+                td[gi].copy_(clamped)
 
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):
         """Implicit-eigen preconditioning on every rank (reference
         :137-144), same-shape layers batched, replayed as one hipGraph
         (launch-bound phase; see base._run_graphed)."""
-        mods = self.modules
+        from kfac_pytorch_amd.ops.factors import factor_groups
+        from kfac_pytorch_amd.ops.linalg import eigen_precondition_grouped
+        mods = [m for m in self.modules if factor_groups(m) == 1]
+        gmods = [m for m in self.modules if factor_groups(m) > 1]
 
         def fn():
             preds = eigen_precondition_multi(
@@ -266,21 +285,26 @@ class KFACEigen(EigenComputeMixin, KFACInverse):
                 [self.m_QG[m] for m in mods],
                 [self.m_dG[m] for m in mods],
                 [self._get_grad(m) for m in mods], self.damping)
-            for m, p in zip(mods, preds):
+            for m in gmods:
+                preds.append(eigen_precondition_grouped(
+                    self.m_QA[m], self.m_dA[m], self.m_QG[m],
+                    self.m_dG[m], self._get_grad(m), self.damping))
+            for m, p in zip(mods + gmods, preds):
                 if m in self.m_precon_grad and \
                         self.m_precon_grad[m].shape == p.shape:
                     self.m_precon_grad[m].copy_(p)
                 else:
                     self.m_precon_grad[m] = p
 
-        if any(m not in self.m_precon_grad for m in mods):
+        if any(m not in self.m_precon_grad for m in mods + gmods):
             # first step allocates the output buffers eagerly so the
             # captured graph only ever writes into static storage
             fn()
             return
+        allm = mods + gmods
         fp = (float(self.damping),
-              tuple(m.weight.grad.data_ptr() for m in mods),
-              tuple(m.bias.grad.data_ptr() for m in mods
+              tuple(m.weight.grad.data_ptr() for m in allm),
+              tuple(m.bias.grad.data_ptr() for m in allm
                     if m.bias is not None),
-              tuple(self.m_QA[m].data_ptr() for m in mods))
+              tuple(self.m_QA[m].data_ptr() for m in allm))
         self._run_graphed("pred", fn, fp)

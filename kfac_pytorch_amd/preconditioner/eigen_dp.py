@@ -52,11 +52,15 @@ class KFACEigenDP(EigenComputeMixin, KFACInverseDP):
             rank_a, rank_g = self.module_ranks[m]
             da, dg = factor_dims(m)
             if rank == rank_a:
-                self.m_QA[m] = torch.zeros(da, da, device=dev)
-                self.m_dA[m] = torch.zeros(da, device=dev)
+                self.m_QA[m] = torch.zeros(self._block_shape(m, da),
+                                           device=dev)
+                self.m_dA[m] = torch.zeros(
+                    self._block_shape(m, da, vec=True), device=dev)
             if rank == rank_g:
-                self.m_QG[m] = torch.zeros(dg, dg, device=dev)
-                self.m_dG[m] = torch.zeros(dg, device=dev)
+                self.m_QG[m] = torch.zeros(self._block_shape(m, dg),
+                                           device=dev)
+                self.m_dG[m] = torch.zeros(
+                    self._block_shape(m, dg, vec=True), device=dev)
 
     # ------------------------------------------------------------- inverses
     def _compute_inverse(self):
@@ -86,14 +90,23 @@ class KFACEigenDP(EigenComputeMixin, KFACInverseDP):
         if not owned:
             return
 
+        from kfac_pytorch_amd.ops.factors import factor_groups
+        from kfac_pytorch_amd.ops.linalg import eigen_precondition_grouped
+        plain = [m for m in owned if factor_groups(m) == 1]
+        gmods = [m for m in owned if factor_groups(m) > 1]
+
         def fn():
             preds = eigen_precondition_multi(
-                [self.m_QA[m] for m in owned],
-                [self.m_dA[m] for m in owned],
-                [self.m_QG[m] for m in owned],
-                [self.m_dG[m] for m in owned],
-                [self._get_grad(m) for m in owned], self.damping)
-            for m, p in zip(owned, preds):
+                [self.m_QA[m] for m in plain],
+                [self.m_dA[m] for m in plain],
+                [self.m_QG[m] for m in plain],
+                [self.m_dG[m] for m in plain],
+                [self._get_grad(m) for m in plain], self.damping)
+            for m in gmods:
+                preds.append(eigen_precondition_grouped(
+                    self.m_QA[m], self.m_dA[m], self.m_QG[m],
+                    self.m_dG[m], self._get_grad(m), self.damping))
+            for m, p in zip(plain + gmods, preds):
                 self.m_precon_grad[m].copy_(p)
 
         # the ~300 small launches of the grouped-bmm pred phase replay

@@ -64,19 +64,22 @@ class KFACInverse(KFACBase):
         for i, m in enumerate(self.modules):
             rank_a, rank_g = self.module_ranks[m]
             da, dg = factor_dims(m)
-            specs.append((f"invA{i}", (da, da), rank_a))
-            specs.append((f"invG{i}", (dg, dg), rank_g))
+            specs.append((f"invA{i}", self._block_shape(m, da), rank_a))
+            specs.append((f"invG{i}", self._block_shape(m, dg), rank_g))
         self.inv_buckets = self._alloc_owner_buckets(specs)
         for i, m in enumerate(self.modules):
             self.m_inv_A[m] = self._owner_view(self.inv_buckets, f"invA{i}")
             self.m_inv_G[m] = self._owner_view(self.inv_buckets, f"invG{i}")
 
     def _alloc_pred_buckets(self):
+        from kfac_pytorch_amd.ops.factors import factor_groups
         specs = []
         for i, m in enumerate(self.modules):
             rank_a, _ = self.module_ranks[m]
             da, dg = factor_dims(m)
-            specs.append((f"pred{i}", (dg, da), rank_a))
+            gr = factor_groups(m)
+            shape = (gr, dg, da) if gr > 1 else (dg, da)
+            specs.append((f"pred{i}", shape, rank_a))
         self.pred_buckets = self._alloc_owner_buckets(specs)
         for i, m in enumerate(self.modules):
             self.m_precon_grad[m] = self._owner_view(
@@ -87,9 +90,11 @@ class KFACInverse(KFACBase):
             dev = self._state_device()
             rank_a, rank_g = self.module_ranks[m]
             if self.comm.rank() == rank_a:
-                self.m_inv_A[m] = torch.zeros(da, da, device=dev)
+                self.m_inv_A[m] = torch.zeros(
+                    self._block_shape(m, da), device=dev)
             if self.comm.rank() == rank_g:
-                self.m_inv_G[m] = torch.zeros(dg, dg, device=dev)
+                self.m_inv_G[m] = torch.zeros(
+                    self._block_shape(m, dg), device=dev)
 
     # -------------------------------------------------------------- factors
     def _compute_factors(self):
@@ -121,10 +126,11 @@ class KFACInverse(KFACBase):
 
     # ------------------------------------------------------------- inverses
     def _pi_damping(self, m) -> torch.Tensor:
-        """pi = sqrt((trA/dimA)/(trG/dimG)) (reference :121)."""
+        """pi = sqrt((trA/dimA)/(trG/dimG)) (reference :121); for
+        grouped convs one pi per block, shape (g,)."""
         A, G = self.m_A[m], self.m_G[m]
-        trA = A.diagonal().sum() / A.shape[0]
-        trG = G.diagonal().sum() / G.shape[0]
+        trA = A.diagonal(dim1=-2, dim2=-1).sum(-1) / A.shape[-1]
+        trG = G.diagonal(dim1=-2, dim2=-1).sum(-1) / G.shape[-1]
         return torch.sqrt(trA / trG)
 
     def _compute_inverse(self):
@@ -135,20 +141,28 @@ class KFACInverse(KFACBase):
         from kfac_pytorch_amd.ops.linalg import mat_inv_multi
         rank = self.comm.rank()
         sqrt_damp = self.damping ** 0.5
+        from kfac_pytorch_amd.ops.factors import factor_groups
         mats, damps, dests = [], [], []
         for m in self.modules:
             rank_a, rank_g = self.module_ranks[m]
             if rank != rank_a and rank != rank_g:
                 continue
             pi = self._pi_damping(m)
-            if rank == rank_a:
-                mats.append(self.m_A[m])
-                damps.append(sqrt_damp * pi)
-                dests.append(self.m_inv_A[m])
-            if rank == rank_g:
-                mats.append(self.m_G[m])
-                damps.append(sqrt_damp / pi)
-                dests.append(self.m_inv_G[m])
+            gr = factor_groups(m)
+            for gi in (range(gr) if gr > 1 else (None,)):
+                pig = pi if gi is None else pi[gi]
+                if rank == rank_a:
+                    mats.append(self.m_A[m] if gi is None
+                                else self.m_A[m][gi])
+                    damps.append(sqrt_damp * pig)
+                    dests.append(self.m_inv_A[m] if gi is None
+                                 else self.m_inv_A[m][gi])
+                if rank == rank_g:
+                    mats.append(self.m_G[m] if gi is None
+                                else self.m_G[m][gi])
+                    damps.append(sqrt_damp / pig)
+                    dests.append(self.m_inv_G[m] if gi is None
+                                 else self.m_inv_G[m][gi])
         for inv, dst in zip(mat_inv_multi(mats, damp_diag=damps), dests):
             dst.copy_(inv)
 

@@ -96,6 +96,7 @@ class KFACInverseDP(KFACInverse):
         """Owner inverts its own locally-built factors (reference
         :98-123), all of them issued as one pool-overlapped potrf+potri
         batch on GPU (mat_inv_multi)."""
+        from kfac_pytorch_amd.ops.factors import factor_groups
         from kfac_pytorch_amd.ops.linalg import mat_inv_multi
         rank = self.comm.rank()
         sqrt_damp = self.damping ** 0.5
@@ -107,15 +108,25 @@ class KFACInverseDP(KFACInverse):
             if rank_a == rank_g and rank == rank_a:
                 pi = self._pi_damping(m)
             else:
-                pi = 1.0
-            if rank == rank_a:
-                mats.append(self.m_A[m])
-                damps.append(sqrt_damp * pi)
-                dests.append(self.m_inv_A[m])
-            if rank == rank_g:
-                mats.append(self.m_G[m])
-                damps.append(sqrt_damp / pi)
-                dests.append(self.m_inv_G[m])
+                pi = None
+            gr = factor_groups(m)
+            for gi in (range(gr) if gr > 1 else (None,)):
+                if pi is None:
+                    pig = 1.0
+                else:
+                    pig = pi if gi is None else pi[gi]
+                if rank == rank_a:
+                    mats.append(self.m_A[m] if gi is None
+                                else self.m_A[m][gi])
+                    damps.append(sqrt_damp * pig)
+                    dests.append(self.m_inv_A[m] if gi is None
+                                 else self.m_inv_A[m][gi])
+                if rank == rank_g:
+                    mats.append(self.m_G[m] if gi is None
+                                else self.m_G[m][gi])
+                    damps.append(sqrt_damp / pig)
+                    dests.append(self.m_inv_G[m] if gi is None
+                                 else self.m_inv_G[m][gi])
         for inv, dst in zip(mat_inv_multi(mats, damp_diag=damps), dests):
             dst.copy_(inv)
 

@@ -42,16 +42,15 @@ def main():
         cls = m.__class__.__name__
         if cls not in ("Linear", "Conv2d"):
             continue
-        if cls == "Conv2d" and m.groups != 1:
-            skipped_groups += 1  # grouped convs are not preconditioned
-            continue
         da, dg = factor_dims(m)
+        from kfac_pytorch_amd.ops.factors import factor_groups
+        if factor_groups(m) > 1:
+            skipped_groups += 1  # counted: grouped = block factors
         rows.append((cls, da, dg))
 
     print(f"{args.model}: {len(rows)} hooked layers"
-          + (f" ({skipped_groups} grouped convs skipped -- their factor"
-             " math does not match the weight layout; they get plain SGD"
-             " updates)" if skipped_groups else ""))
+          + (f" ({skipped_groups} grouped convs with exact per-group"
+             " BLOCK factors)" if skipped_groups else ""))
     print(f"{'layer':8s} {'dim A':>6s} {'dim G':>6s} "
           f"{'A bytes':>10s} {'G bytes':>10s} {'eig bcast':>10s}")
     totA = totG = tot_eig = 0

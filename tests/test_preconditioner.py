@@ -214,20 +214,22 @@ def test_dp_kfac_factory(single_process_comm):
                       kfac.KFAC_INV_DP)
 
 
-def test_grouped_conv_is_skipped(single_process_comm):
-    class G(nn.Module):
-        def __init__(self):
-            super().__init__()
-            self.c = nn.Conv2d(4, 4, 3, groups=2, padding=1)
-            self.f = nn.Linear(4, 2)
-
-        def forward(self, x):
-            return self.f(self.c(x).mean((2, 3)))
-
-    model = G()
-    pre = kfac.KFAC_EIGEN_DP(model)
-    assert len(pre.modules) == 1  # only the Linear
-
+def test_grouped_conv_is_hooked_with_block_factors(single_process_comm,
+                                                   seeded):
+    """Grouped convs are preconditioned with per-group block factors
+    (round 2; full oracle in tests/test_grouped_conv.py)."""
+    import torch.nn as nn
+    model = nn.Sequential(nn.Conv2d(4, 8, 3, groups=2, padding=1),
+                          nn.Flatten(), nn.Linear(8 * 4 * 4, 3))
+    pre = kfac.KFAC_EIGEN_DP(model, damping=0.01)
+    assert len(pre.modules) == 2
+    x = torch.randn(2, 4, 4, 4)
+    y = torch.randint(0, 3, (2,))
+    model.zero_grad(set_to_none=False)
+    torch.nn.functional.cross_entropy(model(x), y).backward()
+    pre.step()
+    gc = model[0]
+    assert pre.m_A[gc].shape[0] == 2  # one block per group
 
 def test_kfac_state_dict_roundtrip(single_process_comm, seeded):
     """Warm-resume: save K-FAC state, rebuild a fresh preconditioner,
