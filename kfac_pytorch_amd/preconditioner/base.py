@@ -226,14 +226,22 @@ class KFACBase(optim.Optimizer):
             entry[0].replay()
             return
         try:
+            import warnings
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
                 fn()  # warmup (allocator state, lazy inits)
             torch.cuda.current_stream().wait_stream(side)
             graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                fn()
+            with warnings.catch_warnings(record=True) as caught:
+                warnings.simplefilter("always")
+                with torch.cuda.graph(graph):
+                    fn()
+            # torch only WARNS when a capture lands empty; an empty
+            # graph would replay as a silent no-op forever -- treat it
+            # as a capture failure and stay eager
+            if any("empty" in str(w.message).lower() for w in caught):
+                raise RuntimeError("empty hipGraph capture")
             self._graphs[key] = (graph, fingerprint)
             graph.replay()
         except Exception:
