@@ -63,6 +63,17 @@ __device__ __forceinline__ float load_sc1(const float* p) {
   return __uint_as_float(__hip_atomic_load(
       (const unsigned*)p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
 }
+// 8-byte sc1 load (two floats): scalar 4-B sc1 traffic costs ~6x the
+// wide-op rate per byte; pairing halves the instruction count and runs
+// at ~0.6x the 16-B rate (guide sc1 price rows)
+__device__ __forceinline__ void load2_sc1(const float* p, float* a,
+                                          float* b) {
+  unsigned long long v = __hip_atomic_load(
+      (const unsigned long long*)p, __ATOMIC_RELAXED,
+      __HIP_MEMORY_SCOPE_AGENT);
+  *a = __uint_as_float((unsigned)v);
+  *b = __uint_as_float((unsigned)(v >> 32));
+}
 
 __device__ __forceinline__ float wave_reduce(float v) {
 #pragma unroll
@@ -312,10 +323,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       const float* Aj = A_m + (size_t)j * n;
       for (int t4 = j0 + 4 * tid; t4 < n; t4 += 1024) {
         float4 v;
-        v.x = load_sc1(Aj + t4);
-        v.y = load_sc1(Aj + t4 + 1);
-        v.z = load_sc1(Aj + t4 + 2);
-        v.w = load_sc1(Aj + t4 + 3);
+        load2_sc1(Aj + t4, &v.x, &v.y);
+        load2_sc1(Aj + t4 + 2, &v.z, &v.w);
         *(float4*)(x_lds + (t4 - j0)) = v;
       }
     }
