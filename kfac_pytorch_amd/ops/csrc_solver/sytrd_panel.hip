@@ -11,7 +11,7 @@
 //   over all same-(padded-)dim factors: grid = (wgs_per_matrix, batch).
 //   All 64 column iterations of the panel run INSIDE the launch with
 //   per-matrix grid barriers (monotonic counter, relaxed sc1 poll +
-//   s_sleep; fence-free -- published data is sc1 both sides) -- 2 per
+//   s_sleep, agent-scope release/acquire fences) -- 2 barriers per
 //   column instead of rocSOLVER's ~4 kernel launches per column; while
 //   one matrix waits at its barrier the other matrices' workgroups
 //   stream their matvecs, so the barrier latency hides in the batch.
@@ -46,35 +46,6 @@
 
 namespace {
 
-// Write-through (sc1) scalar store/load: the cross-workgroup data of
-// this kernel (the working x row, the W panel columns, the reduction
-// scratch, the scaled reflector rows) is published with sc1 stores and
-// read with sc1 loads (guide G16 valid form "sc1 both sides"), which
-// removes BOTH barrier fences: no L2 write-back on release, no L1
-// invalidate on acquire.  Plain loads stay only where a value cannot
-// have a stale L1 copy (the big trailing-matrix streams, own-thread
-// round trips) -- note the panel rows ARE matvec-cached pre-correction
-// within a panel, so their cross-phase reads must be sc1.
-__device__ __forceinline__ void store_sc1(float* p, float v) {
-  __hip_atomic_store((unsigned*)p, __float_as_uint(v), __ATOMIC_RELAXED,
-                     __HIP_MEMORY_SCOPE_AGENT);
-}
-__device__ __forceinline__ float load_sc1(const float* p) {
-  return __uint_as_float(__hip_atomic_load(
-      (const unsigned*)p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT));
-}
-// 8-byte sc1 load (two floats): scalar 4-B sc1 traffic costs ~6x the
-// wide-op rate per byte; pairing halves the instruction count and runs
-// at ~0.6x the 16-B rate (guide sc1 price rows)
-__device__ __forceinline__ void load2_sc1(const float* p, float* a,
-                                          float* b) {
-  unsigned long long v = __hip_atomic_load(
-      (const unsigned long long*)p, __ATOMIC_RELAXED,
-      __HIP_MEMORY_SCOPE_AGENT);
-  *a = __uint_as_float((unsigned)v);
-  *b = __uint_as_float((unsigned)(v >> 32));
-}
-
 __device__ __forceinline__ float wave_reduce(float v) {
 #pragma unroll
   for (int off = 32; off; off >>= 1) v += __shfl_down(v, off);
@@ -85,12 +56,15 @@ __device__ __forceinline__ float wave_reduce(float v) {
 // Returns false when the matrix has been aborted (status set).
 __device__ bool grid_barrier(unsigned* cnt, int* status, int wgs,
                              unsigned phase) {
-  // every wave drains its outstanding (write-through) stores, then one
-  // lane arrives; consumers read published data with sc1 loads, so no
-  // release/acquire fences are needed (guide G16 "sc1 both sides")
+  // every storing wave drains its own outstanding global stores BEFORE
+  // the leader's release fence (guide Appendix A hand-off checklist)
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   if (threadIdx.x == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    // the compiler may drop the vmcnt wait behind buffer_wbl2 when its
+    // scoreboard is provably empty; restate it (guide G16 pitfall 12)
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __hip_atomic_fetch_add(cnt, 1u, __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_AGENT);
     const unsigned target = phase * (unsigned)wgs;
@@ -115,6 +89,7 @@ __device__ bool grid_barrier(unsigned* cnt, int* status, int wgs,
         }
       }
     }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
   __syncthreads();
   // abort propagates: everyone re-checks status after the fence
@@ -184,7 +159,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       // alpha of the previous column from the pwv partials
       float p = 0.f;
       for (int t = tid; t < wgs; t += 256)
-        p += load_sc1(&scr[(size_t)129 * wgs_alloc + t]);
+        p += scr[(size_t)129 * wgs_alloc + t];
       p = wave_reduce(p);
       if (lane == 0) red[wave] = p;
       __syncthreads();
@@ -192,15 +167,12 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         alpha[i - 1] = -0.5f * tau_prev * (red[0] + red[1] +
                                            red[2] + red[3]);
       __syncthreads();
-      // scaled-v writeback of column j-1 + slab append (own rows).
-      // row j-1 was matvec-streamed (plain, L1-cached) at earlier
-      // columns of this panel, so its corrected values must come via
-      // sc1 loads
+      // scaled-v writeback of column j-1 + slab append (own rows)
       float* Aprev = A_m + (size_t)(j - 1) * n;
       float* Wprev = W_m + (size_t)(i - 1) * n;
       for (int r = max(r0, j) + tid; r < r1; r += 256) {
-        float v = (r == j) ? 1.f : s_prev * load_sc1(&Aprev[r]);
-        store_sc1(&Aprev[r], v);
+        float v = (r == j) ? 1.f : s_prev * Aprev[r];
+        Aprev[r] = v;
         slabV[(i - 1) * RS + (r - r0)] = v;
         slabW[(i - 1) * RS + (r - r0)] = Wprev[r];
       }
@@ -211,9 +183,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       if (wave == 0 && lane < i) {
         float cb = (lane == i - 1)
                        ? 1.f
-                       : load_sc1(&A_m[(size_t)(j0 + lane) * n + j]);
-        coefA[lane] = load_sc1(&W_m[(size_t)lane * n + j]) +
-                      2.f * alpha[lane] * cb;
+                       : A_m[(size_t)(j0 + lane) * n + j];
+        coefA[lane] = W_m[(size_t)lane * n + j] + 2.f * alpha[lane] * cb;
         coefB[lane] = cb;
       }
     }
@@ -230,9 +201,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
                  slabW[lane * RS + rl] * coefB[lane];
         corr = wave_reduce(corr);
         if (lane == 0) {
-          // row j was matvec-streamed at earlier columns: sc1 round trip
-          float x = load_sc1(&Aj[r]) - corr;
-          store_sc1(&Aj[r], x);
+          float x = Aj[r] - corr;
+          Aj[r] = x;
           x_own[rl] = x;
         }
       }
@@ -253,7 +223,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         if (!isfinite(t0))
           __hip_atomic_store(st, 2, __ATOMIC_RELAXED,
                              __HIP_MEMORY_SCOPE_AGENT);
-        store_sc1(&scr[(size_t)0 * wgs_alloc + w], t0);
+        scr[(size_t)0 * wgs_alloc + w] = t0;
       }
       // pV/pW: waves split the c-range, lanes... lanes = c, loop rows.
       const int rlo = max(r0, j + 1);
@@ -267,8 +237,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         pv = wave_reduce(pv);
         pw = wave_reduce(pw);
         if (lane == 0) {
-          store_sc1(&scr[(size_t)(1 + c) * wgs_alloc + w], pv);
-          store_sc1(&scr[(size_t)(65 + c) * wgs_alloc + w], pw);
+          scr[(size_t)(1 + c) * wgs_alloc + w] = pv;
+          scr[(size_t)(65 + c) * wgs_alloc + w] = pw;
         }
       }
     }
@@ -283,14 +253,14 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
         if (live) {
           float acc = 0.f;
           for (int t = 0; t < wgs; ++t)
-            acc += load_sc1(&scr[(size_t)s * wgs_alloc + t]);
+            acc += scr[(size_t)s * wgs_alloc + t];
           sums[s] = acc;
         }
       }
       __syncthreads();
     }
     // larfg scalars, redundant per thread
-    const float x1 = load_sc1(&A_m[(size_t)j * n + j + 1]);
+    const float x1 = A_m[(size_t)j * n + j + 1];
     const float nrm2 = sums[0];
     float beta, tau_j, s;
     if (nrm2 == 0.f) {
@@ -309,8 +279,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
     }
     // matvec correction coefficients (one wave, lane = c)
     if (wave == 0 && lane < i) {
-      float vj1 = load_sc1(&A_m[(size_t)(j0 + lane) * n + j + 1]);
-      float wj1 = load_sc1(&W_m[(size_t)lane * n + j + 1]);
+      float vj1 = A_m[(size_t)(j0 + lane) * n + j + 1];
+      float wj1 = W_m[(size_t)lane * n + j + 1];
       float sV = s * (sums[1 + lane] - vj1 * x1) + vj1;
       float sW = s * (sums[65 + lane] - wj1 * x1) + wj1;
       qA[lane] = sW + 2.f * alpha[lane] * sV;
@@ -321,12 +291,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
     // issue in the hot loop (per-CU streaming is issue-bound)
     {
       const float* Aj = A_m + (size_t)j * n;
-      for (int t4 = j0 + 4 * tid; t4 < n; t4 += 1024) {
-        float4 v;
-        load2_sc1(Aj + t4, &v.x, &v.y);
-        load2_sc1(Aj + t4 + 2, &v.z, &v.w);
-        *(float4*)(x_lds + (t4 - j0)) = v;
-      }
+      for (int t4 = j0 + 4 * tid; t4 < n; t4 += 1024)
+        *(float4*)(x_lds + (t4 - j0)) = *(const float4*)(Aj + t4);
     }
     __syncthreads();
     // the trailing matvec: each wave streams FOUR rows concurrently
@@ -388,7 +354,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
           contrib = wave_reduce(contrib);
           if (lane == 0) {
             float w1 = tau_j * contrib;
-            store_sc1(&W_m[(size_t)i * n + r], w1);
+            W_m[(size_t)i * n + r] = w1;
             float vt = (r == j + 1) ? 1.f : s * x_own[rl];
             pwv_acc += w1 * vt;
           }
@@ -398,8 +364,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       if (lane == 0) red[wave] = pwv_acc;
       __syncthreads();
       if (tid == 0)
-        store_sc1(&scr[(size_t)129 * wgs_alloc + w],
-                  red[0] + red[1] + red[2] + red[3]);
+        scr[(size_t)129 * wgs_alloc + w] =
+            red[0] + red[1] + red[2] + red[3];
     }
     tau_prev = tau_j;
     s_prev = s;
@@ -410,7 +376,7 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
   {
     float p = 0.f;
     for (int t = tid; t < wgs; t += 256)
-      p += load_sc1(&scr[(size_t)129 * wgs_alloc + t]);
+      p += scr[(size_t)129 * wgs_alloc + t];
     p = wave_reduce(p);
     if (lane == 0) red[wave] = p;
     __syncthreads();
@@ -421,8 +387,8 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
     const int jl = j0 + ib - 1;
     float* Al = A_m + (size_t)jl * n;
     for (int r = max(r0, jl + 1) + tid; r < r1; r += 256) {
-      float v = (r == jl + 1) ? 1.f : s_prev * load_sc1(&Al[r]);
-      store_sc1(&Al[r], v);
+      float v = (r == jl + 1) ? 1.f : s_prev * Al[r];
+      Al[r] = v;
     }
     __syncthreads();
     for (int c = 0; c < ib; ++c) {
@@ -431,12 +397,10 @@ __global__ __launch_bounds__(256) void latrd_panel_kernel(
       float* Wc = W_m + (size_t)c * n;
       if (c == ib - 1) {
         for (int r = max(r0, jc + 1) + tid; r < r1; r += 256)
-          store_sc1(&Wc[r],
-                    load_sc1(&Wc[r]) + a * load_sc1(&Al[r]));
+          Wc[r] += a * Al[r];
       } else {
         for (int r = max(r0, jc + 1) + tid; r < r1; r += 256)
-          store_sc1(&Wc[r],
-                    load_sc1(&Wc[r]) + a * slabV[c * RS + (r - r0)]);
+          Wc[r] += a * slabV[c * RS + (r - r0)];
       }
     }
   }
