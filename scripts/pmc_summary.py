@@ -15,7 +15,10 @@ with open(path) as fh:
     disp = defaultdict(set)
     dispcol = next((c for c in cols if "Dispatch" in c and "Id" in c), None)
     for row in r:
-        name = row[namecol].split("(")[0].strip('"')
+        raw = row[namecol].strip('"')
+        if raw.startswith("(anonymous namespace)::"):
+            raw = raw[len("(anonymous namespace)::"):]
+        name = raw.split("(")[0].strip() or raw[:55]
         agg[name][row[ctrcol]] += float(row[valcol])
         if dispcol:
             disp[name].add(row[dispcol])
@@ -25,9 +28,10 @@ for name, ctrs in sorted(agg.items(),
                          key=lambda kv: -kv[1].get("GRBM_GUI_ACTIVE", 0)):
     mfma = ctrs.get("SQ_VALU_MFMA_BUSY_CYCLES", 0)
     gui = ctrs.get("GRBM_GUI_ACTIVE", 0)
-    # GUI_ACTIVE counts chip cycles per dispatch; MFMA busy counts
-    # per-SIMD cycles summed over 1024 SIMDs -> utilization =
-    # MFMA_BUSY / (GUI_ACTIVE * 1024)
+    # RELATIVE MFMA-occupancy proxy: SQ_VALU_MFMA_BUSY summed over all
+    # SIMDs vs chip-active cycles x 1024 SIMDs.  gfx950 ships no
+    # derived-counter formulas (guide: rocprofv3 PMC slots), so compare
+    # kernels against each other, not against an absolute peak.
     util = mfma / (gui * 1024) * 100 if gui else 0.0
     print(f"{name[:55]:55s} {len(disp[name]):>10d} {mfma:14.3e} "
           f"{ctrs.get('SQ_WAVE_CYCLES', 0):14.3e} "
