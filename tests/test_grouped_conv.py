@@ -166,3 +166,27 @@ def test_mobilenet_depthwise_now_preconditioned(single_comm):
     m0 = depthwise[0]
     assert pre.m_A[m0].dim() == 3
     assert torch.isfinite(m0.weight.grad).all()
+
+
+def test_grouped_state_checkpoints(single_comm):
+    """kfac_state_dict round-trips the 3-D grouped factor state."""
+    import kfac_pytorch_amd as kfac
+    torch.manual_seed(5)
+    gm = _GroupedNet()
+    pre = kfac.KFAC_EIGEN_DP(gm, damping=0.01)
+    x = torch.randn(4, 3, 4, 4)
+    y = torch.randint(0, 5, (4,))
+    gm.zero_grad(set_to_none=False)
+    F.cross_entropy(gm(x), y).backward()
+    pre.step()
+    from kfac_pytorch_amd.preconditioner.base import (
+        kfac_state_dict, load_kfac_state_dict)
+    sd = kfac_state_dict(pre)
+    gm2 = _GroupedNet()
+    pre2 = kfac.KFAC_EIGEN_DP(gm2, damping=0.01)
+    gm2.zero_grad(set_to_none=False)
+    F.cross_entropy(gm2(x), y).backward()
+    pre2.step()  # allocate state
+    load_kfac_state_dict(pre2, sd)
+    torch.testing.assert_close(pre2.m_A[gm2.gc], pre.m_A[gm.gc])
+    assert pre2.m_A[gm2.gc].dim() == 3
