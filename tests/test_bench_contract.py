@@ -63,3 +63,28 @@ def test_bench_world4_gloo_end_to_end():
     # whole-job aggregate: 4 ranks x bs 2 x 2 steps
     assert d["config"]["global_batch"] == 8
     assert d["value"] > 0
+
+
+def test_bench_world2_gloo_mpd_eigen():
+    """The comm-heavy MPD 'eigen' algorithm (factor allreduce +
+    eigenbasis broadcasts on rotating groups) through bench.py's own
+    launch path at world 2 -- the SCALE run flips --kfac-name eigen."""
+    env = dict(os.environ)
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+              "MASTER_PORT"):
+        env.pop(k, None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29747", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--batch-size", "2", "--image-size", "64",
+         "--model", "resnet18", "--kfac-name", "eigen",
+         "--dtype", "fp32"],
+        capture_output=True, text=True, timeout=900, cwd=REPO, env=env)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    json_lines = [ln for ln in r.stdout.splitlines()
+                  if ln.startswith("{")]
+    d = json.loads(json_lines[0])
+    assert d["config"]["kfac"] == "eigen" and d["n_gpus"] == 2
+    assert d["value"] > 0
