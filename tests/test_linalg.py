@@ -120,3 +120,19 @@ def test_eigen_clamp_matches_reference_formula():
     clamped = d * (d > eps)
     assert torch.equal(clamped,
                        torch.tensor([0.0, 0.0, 0.0, 1e-9, 0.5]))
+
+
+def test_deferred_info_raises_once():
+    """Nonzero rocSOLVER info words accumulate device-side and raise at
+    the next check, then clear (the sync-free validation path)."""
+    import pytest
+    from kfac_pytorch_amd.ops.linalg import (_defer_info,
+                                             check_deferred_info)
+    check_deferred_info()  # clean state
+    _defer_info(torch.tensor([0, 0, 0], dtype=torch.int32))
+    check_deferred_info()  # zeros: no raise
+    _defer_info(torch.tensor([0, 3, 0], dtype=torch.int32))
+    _defer_info(torch.tensor([1], dtype=torch.int32))
+    with pytest.raises(RuntimeError, match="2 rocSOLVER"):
+        check_deferred_info()
+    check_deferred_info()  # flag cleared by the raise
