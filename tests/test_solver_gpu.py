@@ -232,3 +232,19 @@ def test_custom_sytrd_precondition_accuracy(solver, monkeypatch):
              @ Q_ref.mT)
     err = ((P - P_ref).norm() / P_ref.norm()).item()
     assert err < 5e-3, err
+
+
+def test_mat_inv_multi_trsm_tier(solver):
+    """The GEMM-rate inverse tier (chol + trsm + X^T X) for buckets
+    >= 512 must match torch.cholesky_inverse exactly, padded buckets
+    and singletons included."""
+    from kfac_pytorch_amd.ops.linalg import mat_inv_multi
+    dims = [2304, 2049, 2048, 1152, 700, 512, 256, 256, 64]
+    mats = [spd(m, seed=m + 3) for m in dims]
+    out = mat_inv_multi(mats, damp_diag=[0.01] * len(mats))
+    torch.cuda.synchronize()
+    for a, inv in zip(mats, out):
+        damped = a + 0.01 * torch.eye(a.shape[0], device="cuda")
+        ref = torch.cholesky_inverse(torch.linalg.cholesky(damped))
+        err = ((inv - ref).norm() / ref.norm()).item()
+        assert err < 1e-3, (a.shape[0], err)
