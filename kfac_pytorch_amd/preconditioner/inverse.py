@@ -171,21 +171,45 @@ class KFACInverse(KFACBase):
 
     # ----------------------------------------------------------------- pred
     def _compute_pred(self):
+        """Per-layer inv_G @ grad @ inv_A; launch-bound, so the whole
+        phase replays as one captured hipGraph (base._run_graphed)."""
         if self.communicate_inverse_or_not:
             # every rank preconditions every layer with broadcast inverses
-            for m in self.modules:
-                grad = self._get_grad(m)
-                self.m_precon_grad[m] = inverse_precondition(
-                    self.m_inv_A[m], self.m_inv_G[m], grad)
+            mods = self.modules
+
+            def fn():
+                for m in mods:
+                    grad = self._get_grad(m)
+                    p = inverse_precondition(
+                        self.m_inv_A[m], self.m_inv_G[m], grad)
+                    if m in self.m_precon_grad and \
+                            self.m_precon_grad[m].shape == p.shape:
+                        self.m_precon_grad[m].copy_(p)
+                    else:
+                        self.m_precon_grad[m] = p
+
+            if any(m not in self.m_precon_grad for m in mods):
+                fn()  # first step allocates static output storage
+                return
         else:
             # owner-only pred, broadcast after (reference default :41)
             rank = self.comm.rank()
-            for m in self.modules:
-                rank_a, _ = self.module_ranks[m]
-                if rank == rank_a:
+            mods = [m for m in self.modules
+                    if rank == self.module_ranks[m][0]]
+            if not mods:
+                return
+
+            def fn():
+                for m in mods:
                     grad = self._get_grad(m)
                     self.m_precon_grad[m].copy_(inverse_precondition(
                         self.m_inv_A[m], self.m_inv_G[m], grad))
+
+        fp = (tuple(m.weight.grad.data_ptr() for m in mods),
+              tuple(m.bias.grad.data_ptr() for m in mods
+                    if m.bias is not None),
+              tuple(self.m_inv_A[m].data_ptr() for m in mods))
+        self._run_graphed("pred", fn, fp)
 
     def _communicate_pred(self):
         self._broadcast_owner_buckets(self.pred_buckets)

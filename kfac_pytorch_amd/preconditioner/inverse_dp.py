@@ -134,13 +134,27 @@ class KFACInverseDP(KFACInverse):
     def _compute_pred(self):
         """Owner preconditions the globally-averaged gradient
         (reference :126-138); non-owner bucket views stay zero as the
-        broadcast destination."""
+        broadcast destination.  Launch-bound phase -> one captured
+        hipGraph (base._run_graphed)."""
         assert not self.communicate_inverse_or_not
         rank = self.comm.rank()
+        owned = []
         for m in self.modules:
             rank_a, rank_g = self.module_ranks[m]
             assert rank_a == rank_g
             if rank == rank_a:
+                owned.append(m)
+        if not owned:
+            return
+
+        def fn():
+            for m in owned:
                 grad = self._get_grad(m)
                 self.m_precon_grad[m].copy_(inverse_precondition(
                     self.m_inv_A[m], self.m_inv_G[m], grad))
+
+        fp = (tuple(m.weight.grad.data_ptr() for m in owned),
+              tuple(m.bias.grad.data_ptr() for m in owned
+                    if m.bias is not None),
+              tuple(self.m_inv_A[m].data_ptr() for m in owned))
+        self._run_graphed("pred", fn, fp)
