@@ -344,7 +344,7 @@ def test_pred_graph_matches_eager():
         os.environ.pop("KFAC_PRED_GRAPH", None)
         return out
 
-    for name in ("eigen_dp", "eigen"):
+    for name in ("eigen_dp", "eigen", "inverse", "inverse_dp"):
         g1 = run("1", name)
         g0 = run("0", name)
         for a, b in zip(g1, g0):
