@@ -354,3 +354,28 @@ def test_kfac_state_dict_includes_scheduler_state(single_process_comm,
     pre2.step()
     for a, b in zip(m1.parameters(), m2.parameters()):
         torch.testing.assert_close(a.grad, b.grad, rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.parametrize("name", ["inverse", "eigen", "inverse_dp",
+                                  "eigen_dp"])
+@pytest.mark.parametrize("parts", [
+    "CommunicateInverse", "ComputeInverse",
+    "CommunicateFactor", "ComputeFactor",
+    "CommunicateInverse,ComputeInverse",
+    "CommunicateFactor,CommunicateInverse",
+    "ComputeFactor,ComputeInverse,CommunicateFactor,CommunicateInverse",
+])
+def test_exclude_parts_config_space(single_process_comm, seeded, name,
+                                    parts):
+    """Every exclude_parts ablation flag combination must run every
+    algorithm without crashing and leave finite gradients (the
+    reference's time-breakdown harness relies on all of these,
+    kfac_preconditioner_base.py:96-99,200-225)."""
+    model = TinyNet()
+    pre = kfac.get_kfac_module(name)(model, damping=0.01,
+                                     exclude_parts=parts)
+    for step in range(2):
+        run_fwd_bwd(model, seed=step)
+        pre.step()
+    for p in model.parameters():
+        assert torch.isfinite(p.grad).all()
