@@ -74,8 +74,8 @@ class Seq2SeqTransformer(nn.Module):
     @torch.no_grad()
     def greedy_decode(self, src, max_len: int = 32, bos: int = 1,
                       eos: int = 2):
-        """Greedy decoding (capability analog of the reference's
-        beam-search Translator, examples/transformer/Translator.py)."""
+        """Greedy decoding (fast path; the evaluation decoder is
+        :meth:`beam_decode`)."""
         self.eval()
         ys = torch.full((src.size(0), 1), bos, dtype=torch.long,
                         device=src.device)
@@ -86,6 +86,52 @@ class Seq2SeqTransformer(nn.Module):
             if (nxt == eos).all():
                 break
         return ys
+
+    @torch.no_grad()
+    def beam_decode(self, src, beam_size: int = 4, max_len: int = 32,
+                    bos: int = 1, eos: int = 2,
+                    length_penalty: float = 0.6):
+        """Beam-search decoding of ONE source sequence (src (1, S) or
+        (S,)) -- the reference's BLEU evaluation decoder
+        (examples/transformer/Translator.py; length-normalized
+        log-probability scoring).  Returns the best (L,) token tensor
+        including bos/eos."""
+        self.eval()
+        if src.dim() == 1:
+            src = src.unsqueeze(0)
+        assert src.size(0) == 1, "beam_decode takes one sequence"
+        dev = src.device
+        beams = torch.full((1, 1), bos, dtype=torch.long, device=dev)
+        scores = torch.zeros(1, device=dev)
+        finished: list = []
+        for _ in range(max_len - 1):
+            nb = beams.size(0)
+            logits = self.forward(src.expand(nb, -1), beams)
+            logp = torch.log_softmax(logits[:, -1].float(), dim=-1)
+            cand = scores.unsqueeze(1) + logp          # (nb, V)
+            flat = cand.reshape(-1)
+            k = min(beam_size, flat.numel())
+            top, idx = flat.topk(k)
+            parent = idx // logp.size(-1)
+            token = idx % logp.size(-1)
+            beams = torch.cat([beams[parent], token.unsqueeze(1)], 1)
+            scores = top
+            done = token == eos
+            for i in torch.nonzero(done).flatten().tolist():
+                norm = float(scores[i]) / (beams.size(1) **
+                                           length_penalty)
+                finished.append((norm, beams[i].clone()))
+            keep = ~done
+            if not bool(keep.any()):
+                break
+            beams, scores = beams[keep], scores[keep]
+            if len(finished) >= beam_size:
+                break
+        if not finished:  # no eos within max_len: best open beam
+            i = int(scores.argmax())
+            return beams[i]
+        finished.sort(key=lambda t: -t[0])
+        return finished[0][1]
 
 
 def make_transformer(vocab: int = 9521, **kw) -> Seq2SeqTransformer:
