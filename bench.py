@@ -72,6 +72,12 @@ def init_dist(args):
     backend = "nccl" if use_cuda else "gloo"
     if use_cuda:
         torch.cuda.set_device(local_rank)
+        # MIOpen find-mode autotuning for the conv fw/bw (the
+        # reference sets cudnn.benchmark=True,
+        # examples/pytorch_cifar10_resnet.py:134); KFAC_CONV_BENCHMARK=0
+        # disables
+        torch.backends.cudnn.benchmark = (
+            os.environ.get("KFAC_CONV_BENCHMARK", "1") != "0")
     dist.init_process_group(backend=backend, init_method="env://",
                             world_size=world, rank=rank)
     return rank, world, local_rank, use_cuda
