@@ -3,7 +3,7 @@
 cd "$(dirname "$0")/.."
 run() {
     name=$1; shift
-    KFAC_PHASE_TIMING=1 timeout 180 python bench.py --steps 10 --warmup 3 "$@" \
+    KFAC_PHASE_TIMING=1 timeout 330 python bench.py --steps 10 --warmup 3 "$@" \
         2>/dev/null | grep -E "^KFAC_PHASES|^\{" | tail -2 \
         | sed "s/^/[$name] /"
 }
