@@ -183,10 +183,12 @@ def get_imagenet_model(name: str, num_classes: int = 1000) -> nn.Module:
     resnext here, densenet/vgg/inception/mobilenet in imagenet_extras."""
     if name not in _MODELS:
         from kfac_pytorch_amd.models import imagenet_extras as ex
+        from kfac_pytorch_amd.models import vit as _vit
         extras = {
             "densenet121": ex.densenet121, "densenet201": ex.densenet201,
             "inceptionv3": ex.inception_v3, "inceptionv4": ex.inception_v4,
             "mobilenetv2": ex.mobilenet_v2, "vgg16": ex.vgg16_imagenet,
+            "vit_tiny": _vit.vit_tiny, "vit_small": _vit.vit_small,
         }
         if name in extras:
             return extras[name](num_classes=num_classes)

@@ -162,3 +162,37 @@ def test_transformer_beam_decode():
         s = sum(float(lp[t, seq[t + 1]]) for t in range(seq.numel() - 1))
         return s / (seq.numel() ** 0.6)
     assert score(out1) >= score(g) - 1e-4
+
+
+def test_vit_kfac_step():
+    """ViT (beyond-reference zoo entry): every parameterized layer is
+    Linear/Conv2d, the full model preconditions end to end."""
+    import torch
+    import torch.nn.functional as F
+    import torch.distributed as dist
+    import kfac_pytorch_amd as kfac
+    import kfac_pytorch_amd.parallel.comm as comm_mod
+    from tests.conftest import free_port
+    if not dist.is_initialized():
+        dist.init_process_group(
+            "gloo", init_method=f"tcp://127.0.0.1:{free_port()}",
+            world_size=1, rank=0)
+    comm_mod.reset()
+    comm_mod.init("Torch")
+    from kfac_pytorch_amd.models import get_imagenet_model
+    torch.manual_seed(0)
+    m = get_imagenet_model("vit_tiny", num_classes=10)
+    # shrink for CPU: 32px, patch 16 -> 4 tokens
+    from kfac_pytorch_amd.models.vit import VisionTransformer
+    m = VisionTransformer(image_size=32, patch=16, dim=48, depth=2,
+                          heads=3, num_classes=10)
+    pre = kfac.KFAC_EIGEN_DP(m, damping=0.01)
+    x = torch.randn(4, 3, 32, 32)
+    y = torch.randint(0, 10, (4,))
+    m.zero_grad(set_to_none=False)
+    F.cross_entropy(m(x), y).backward()
+    pre.step()
+    # patchify conv + 6 linears/block x2 + head = 14 hooked layers
+    assert len(pre.modules) == 1 + 6 * 2 + 1
+    for p in m.parameters():
+        assert torch.isfinite(p.grad).all()
