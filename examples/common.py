@@ -84,6 +84,10 @@ def initialize_distributed():
     use_cuda = torch.cuda.is_available()
     if use_cuda:
         torch.cuda.set_device(local_rank)
+        # MIOpen find-mode conv autotuning (reference trainers set
+        # cudnn.benchmark=True, examples/pytorch_cifar10_resnet.py:134)
+        torch.backends.cudnn.benchmark = (
+            os.environ.get("KFAC_CONV_BENCHMARK", "1") != "0")
     dist.init_process_group("nccl" if use_cuda else "gloo",
                             init_method="env://", rank=rank,
                             world_size=world)
