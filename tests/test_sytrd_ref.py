@@ -48,3 +48,34 @@ def test_blocked_sytrd_diagonal_input():
     assert np.allclose(d, d_in)
     assert np.allclose(e[:31], 0.0)
     assert np.allclose(tau[:31], 0.0)
+
+
+@pytest.mark.parametrize("n,b", [(24, 4), (48, 8), (65, 8)])
+def test_two_stage_sbr_reference(n, b):
+    """Round-3 target algebra (docs/SBR_STAGE2_NOTES.md): full->band
+    via CholeskyQR2 panels, band->tridiagonal via Givens bulge chasing
+    -- both stages exact similarity transforms with the intended
+    structure."""
+    from scripts.sbr_ref import self_check
+    r = self_check(n, b, seed=n)
+    assert r["band_resid"] < 1e-12
+    assert r["tri_resid"] < 1e-12
+    assert r["recon_tri"] < 1e-12
+    assert r["eig_err"] < 1e-12
+
+
+def test_sbr_rank_deficient_panels():
+    """CholeskyQR2's Householder fallback on rank-deficient panels
+    (early-training K-FAC factors are rank-deficient)."""
+    from scripts.sbr_ref import self_check
+    import numpy as np
+    rng = np.random.default_rng(1)
+    n, r = 40, 8
+    x = rng.standard_normal((n, r))
+    a = x @ x.T / r          # rank 8 of 40
+    from scripts.sbr_ref import band_reduce, bulge_chase
+    B, Q1 = band_reduce(a, 4)
+    T, Q2 = bulge_chase(B, 4)
+    Q = Q1 @ Q2
+    assert np.linalg.norm(Q @ T @ Q.T - a) / np.linalg.norm(a) < 1e-11
+    assert np.max(np.abs(np.triu(T, 2))) < 1e-11
