@@ -10,10 +10,11 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_bench_json_contract():
+    from tests.conftest import free_port
     env = dict(os.environ)
     for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK"):
         env.pop(k, None)
-    env["MASTER_PORT"] = "29733"
+    env["MASTER_PORT"] = str(free_port())
     r = subprocess.run(
         [sys.executable, os.path.join(REPO, "bench.py"), "--steps", "1",
          "--warmup", "0", "--batch-size", "2", "--image-size", "64",
@@ -40,6 +41,7 @@ def test_bench_world4_gloo_end_to_end():
     """The driver's multi-GPU launch pattern, on CPU/gloo at world 4:
     torch.distributed.run -> bench.py --gpus 4, rank-0 JSON aggregate.
     Proves the launch/aggregation path the 8x MI355X SCALE run uses."""
+    from tests.conftest import free_port
     env = dict(os.environ)
     for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
               "MASTER_PORT"):
@@ -47,7 +49,7 @@ def test_bench_world4_gloo_end_to_end():
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
-         "--master-port", "29741", os.path.join(REPO, "bench.py"),
+         "--master-port", str(free_port()), os.path.join(REPO, "bench.py"),
          "--gpus", "4", "--steps", "2", "--warmup", "1",
          "--batch-size", "2", "--image-size", "64",
          "--model", "resnet18", "--kfac-name", "inverse_dp",
@@ -69,6 +71,7 @@ def test_bench_world2_gloo_mpd_eigen():
     """The comm-heavy MPD 'eigen' algorithm (factor allreduce +
     eigenbasis broadcasts on rotating groups) through bench.py's own
     launch path at world 2 -- the SCALE run flips --kfac-name eigen."""
+    from tests.conftest import free_port
     env = dict(os.environ)
     for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
               "MASTER_PORT"):
@@ -76,7 +79,7 @@ def test_bench_world2_gloo_mpd_eigen():
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29747", os.path.join(REPO, "bench.py"),
+         "--master-port", str(free_port()), os.path.join(REPO, "bench.py"),
          "--gpus", "2", "--steps", "2", "--warmup", "1",
          "--batch-size", "2", "--image-size", "64",
          "--model", "resnet18", "--kfac-name", "eigen",
