@@ -127,3 +127,16 @@ def test_dp_block_partition_cli():
     r = _run([os.path.join(SCRIPTS, "dp_block_partition.py")])
     assert r.returncode == 0, r.stderr[-500:]
     assert "block" in r.stdout.lower() or "partition" in r.stdout.lower()
+
+
+def test_shell_tier_syntax():
+    """Every launcher/driver shell script must at least parse
+    (bash -n): launch_torch.sh, train_*.sh, batch*.sh, smoke script."""
+    import glob
+    scripts = (glob.glob(os.path.join(REPO, "*.sh"))
+               + glob.glob(os.path.join(SCRIPTS, "*.sh")))
+    assert len(scripts) >= 8
+    for s in scripts:
+        r = subprocess.run(["bash", "-n", s], capture_output=True,
+                           text=True)
+        assert r.returncode == 0, (s, r.stderr)
