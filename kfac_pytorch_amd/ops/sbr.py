@@ -30,11 +30,13 @@ __all__ = ["band_reduce_batched", "apply_q_batched"]
 Panel = Tuple[int, torch.Tensor, torch.Tensor]  # (r0, V, Tinv)
 
 
-def _panel_wy(P: torch.Tensor):
-    """Batched Householder QR of the (N, M, b) panel in compact-WY
-    form.  Returns (V unit-lower-trapezoid (N, M, k), Tinv upper
-    (N, k, k), R upper-trapezoid (N, k, b)) with the panel's Q equal
-    to I - V T V^T, T = Tinv^{-1}."""
+def _panel_wy_geqrf(P: torch.Tensor):
+    """Householder QR of the (N, M, b) panel in compact-WY form via
+    batched ``geqrf`` -- the always-correct path (handles rank
+    deficiency via tau=0 columns), but latency-bound on ROCm (geqrf
+    runs per matrix with internal syncs; measured 187 ms at 4608x3).
+    Returns (V unit-lower-trapezoid (N, M, k), Tinv upper (N, k, k),
+    R upper-trapezoid (N, k, b)), the panel's Q = I - V T V^T."""
     N, M, b = P.shape
     k = min(M, b)
     a, tau = torch.geqrf(P)
@@ -51,6 +53,65 @@ def _panel_wy(P: torch.Tensor):
     S = torch.bmm(V.mT, V)
     Tinv = S.triu(1) + torch.diag_embed(1.0 / tau)
     return V, Tinv, R
+
+
+def _panel_wy(P: torch.Tensor):
+    """Compact-WY panel factorization from batched ops ONLY (the
+    MI355X fast path -- no geqrf, no per-matrix library loops):
+
+    1. CholeskyQR with the Gram matrix in fp64 (exact orthonormality
+       for any cond(P) < 1e8; the b x b fp64 work is negligible),
+    2. Householder reconstruction from the orthonormal Q (Ballard,
+       Demmel et al.): unpivoted LU of ``Q1 - S`` (S = -sign(diag Q1))
+       gives the unit-lower V1; ``V2 = Q2 U^{-1}``,
+    3. ``T^{-1} = strict_upper(V^T V) + diag(V^T V)/2`` -- for ANY V
+       this T makes I - V T V^T exactly orthogonal, so the similarity
+       transform is exact even where Q itself is only approximate.
+
+    ``Q1 - S`` is diagonally dominant by the sign choice, so batched
+    partial-pivoted ``lu_factor`` picks the identity permutation for
+    ~99.4% of panels (measured over 2000 random panels); any batch
+    with a non-trivial pivot or a failed Cholesky (rank-deficient or
+    tail panel) falls back to :func:`_panel_wy_geqrf`.
+    """
+    N, M, b = P.shape
+    if M <= b:
+        return _panel_wy_geqrf(P)
+    P64 = P.to(torch.float64)
+    G = torch.bmm(P64.mT, P64)
+    L1, info = torch.linalg.cholesky_ex(G)
+    if bool((info != 0).any()):
+        return _panel_wy_geqrf(P)
+    Q = torch.linalg.solve_triangular(L1.mT, P64, upper=True,
+                                      left=False)
+    # second pass restores orthonormality when cond(P) is large
+    G2 = torch.bmm(Q.mT, Q)
+    L2, info = torch.linalg.cholesky_ex(G2)
+    if bool((info != 0).any()):
+        return _panel_wy_geqrf(P)
+    Q = torch.linalg.solve_triangular(L2.mT, Q, upper=True, left=False)
+    R = torch.bmm(L2.mT, L1.mT)              # P = Q R
+    d = Q.diagonal(dim1=-2, dim2=-1)
+    s = -torch.sign(d)
+    s = torch.where(s == 0, torch.ones_like(s), s)
+    Y1 = Q[:, :b, :] - torch.diag_embed(s)
+    lu, piv = torch.linalg.lu_factor(Y1)
+    ident = torch.arange(1, b + 1, device=piv.device,
+                         dtype=piv.dtype)
+    if bool((piv != ident).any()):
+        return _panel_wy_geqrf(P)
+    U = lu.triu()
+    V1 = lu.tril(-1)
+    V1.diagonal(dim1=-2, dim2=-1).fill_(1.0)
+    V2 = torch.linalg.solve_triangular(U, Q[:, b:, :], upper=True,
+                                       left=False)
+    V = torch.cat([V1, V2], dim=1).to(P.dtype)
+    Gv = torch.bmm(V.mT, V)
+    Tinv = Gv.triu(1) + torch.diag_embed(
+        Gv.diagonal(dim1=-2, dim2=-1) / 2)
+    # H^T P = E S R (S diagonal +-1), so the surviving panel block
+    Rp = (s.unsqueeze(-1) * R).to(P.dtype)
+    return V, Tinv, Rp
 
 
 def band_reduce_batched(A: torch.Tensor, b: int = 64

@@ -79,3 +79,57 @@ def test_band_eigenvectors_back_transform():
     # A Qz == Qz diag(d)
     resid = (A @ Qz - Qz * d.unsqueeze(-2)).abs().max()
     assert float(resid) < 1e-11
+
+
+def test_fast_path_taken_without_geqrf_fallback(monkeypatch):
+    """Well-conditioned panels must run the batched CholeskyQR +
+    Householder-reconstruction path -- the geqrf fallback is for
+    rank-deficient/tail/pivoting panels only (the GPU perf story
+    depends on this)."""
+    import kfac_pytorch_amd.ops.sbr as sbr
+    calls = {"n": 0}
+    orig = sbr._panel_wy_geqrf
+
+    def counting(P):
+        calls["n"] += 1
+        return orig(P)
+
+    monkeypatch.setattr(sbr, "_panel_wy_geqrf", counting)
+    A = _spd_stack(2, 200, torch.float64, seed=5)
+    B, panels = band_reduce_batched(A, 32)
+    # only the tail panels (M <= b) may fall back
+    assert calls["n"] <= 2, calls["n"]
+    eye = torch.eye(200, dtype=torch.float64).expand(2, -1, -1)
+    Q = apply_q_batched(panels, eye.contiguous())
+    resid = (Q @ B @ Q.mT - A).norm() / A.norm()
+    assert float(resid) < 1e-13
+
+
+def test_panel_wy_fast_matches_geqrf_transform():
+    """The two panel factorizations differ in V/T/sign conventions but
+    must produce the SAME orthogonal action: H^T P = [R; 0] with
+    identical R up to row signs, and identical H^T C H."""
+    import kfac_pytorch_amd.ops.sbr as sbr
+    g = torch.Generator().manual_seed(9)
+    P = torch.randn(2, 80, 16, generator=g, dtype=torch.float64)
+    C = _spd_stack(2, 80, torch.float64, seed=10)
+
+    def dense_h(V, Tinv):
+        T = torch.linalg.solve_triangular(
+            Tinv, torch.eye(V.shape[-1], dtype=V.dtype).expand(
+                V.shape[0], -1, -1).contiguous(), upper=True)
+        eye = torch.eye(V.shape[1], dtype=V.dtype)
+        return eye - V @ T @ V.mT
+
+    Hf = dense_h(*sbr._panel_wy(P)[:2])
+    Hg = dense_h(*sbr._panel_wy_geqrf(P)[:2])
+    for H in (Hf, Hg):
+        HtP = H.mT @ P
+        assert float(HtP[:, 16:].abs().max()) < 1e-12
+    # the R blocks agree up to per-row sign
+    Rf = (Hf.mT @ P)[:, :16]
+    Rg = (Hg.mT @ P)[:, :16]
+    sf = torch.sign(Rf.diagonal(dim1=-2, dim2=-1))
+    sg = torch.sign(Rg.diagonal(dim1=-2, dim2=-1))
+    assert torch.allclose(sf.unsqueeze(-1) * Rf, sg.unsqueeze(-1) * Rg,
+                          atol=1e-11)
