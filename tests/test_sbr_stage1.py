@@ -133,3 +133,32 @@ def test_panel_wy_fast_matches_geqrf_transform():
     sg = torch.sign(Rg.diagonal(dim1=-2, dim2=-1))
     assert torch.allclose(sf.unsqueeze(-1) * Rf, sg.unsqueeze(-1) * Rg,
                           atol=1e-11)
+
+
+@pytest.mark.parametrize("check", ["deferred", "eager", "geqrf"])
+def test_check_modes_agree(check):
+    """All three failure-handling modes must produce an exact
+    similarity; deferred (the sync-free GPU default) must match the
+    reference geqrf path's band matrix up to roundoff."""
+    A = _spd_stack(2, 150, torch.float64, seed=21)
+    B, panels = band_reduce_batched(A, 32, check=check)
+    eye = torch.eye(150, dtype=torch.float64).expand(2, -1, -1)
+    Q = apply_q_batched(panels, eye.contiguous())
+    resid = (Q @ B @ Q.mT - A).norm() / A.norm()
+    assert float(resid) < 1e-13
+    ev = torch.linalg.eigvalsh(B)
+    ev_ref = torch.linalg.eigvalsh(A)
+    assert float((ev - ev_ref).abs().max() / ev_ref.abs().max()) < 1e-12
+
+
+def test_deferred_redo_on_rank_deficient():
+    """Deferred mode on a rank-deficient stack must detect the flagged
+    panels and redo on the geqrf path -- the result is still an exact
+    similarity (NaNs from the poisoned fast pass must not escape)."""
+    A = _spd_stack(2, 96, torch.float64, seed=3, rank=10)
+    B, panels = band_reduce_batched(A, 16, check="deferred")
+    assert bool(B.isfinite().all())
+    eye = torch.eye(96, dtype=torch.float64).expand(2, -1, -1)
+    Q = apply_q_batched(panels, eye.contiguous())
+    resid = (Q @ B @ Q.mT - A).norm() / A.norm()
+    assert float(resid) < 1e-12
