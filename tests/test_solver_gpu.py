@@ -248,3 +248,22 @@ def test_mat_inv_multi_trsm_tier(solver):
         ref = torch.cholesky_inverse(torch.linalg.cholesky(damped))
         err = ((inv - ref).norm() / ref.norm()).item()
         assert err < 1e-3, (a.shape[0], err)
+
+
+def test_sbr_stage1_gpu_similarity():
+    """SBR stage-1 band reduction (ops/sbr.py) on hardware: the
+    sync-free deferred path at a real bucket dim, fp32 -- similarity
+    and orthogonality at the fp32 tolerance measured in
+    profiles/sbr_stage1_deferred.log (4e-6 at 1152)."""
+    from kfac_pytorch_amd.ops.sbr import (apply_q_batched,
+                                          band_reduce_batched)
+    g = torch.Generator().manual_seed(5)
+    x = torch.randn(2, 576, 1152, generator=g)
+    A = (x @ x.mT / 1152).cuda()
+    B, panels = band_reduce_batched(A, 64, check="deferred")
+    assert float(B.triu(65).abs().max()) == 0.0
+    eye = torch.eye(576, device="cuda").expand(2, -1, -1).contiguous()
+    Q = apply_q_batched(panels, eye)
+    assert float((Q @ Q.mT - eye).abs().max()) < 2e-5
+    resid = (Q @ B @ Q.mT - A).norm() / A.norm()
+    assert float(resid) < 5e-5
