@@ -112,7 +112,6 @@ def test_panel_wy_fast_matches_geqrf_transform():
     import kfac_pytorch_amd.ops.sbr as sbr
     g = torch.Generator().manual_seed(9)
     P = torch.randn(2, 80, 16, generator=g, dtype=torch.float64)
-    C = _spd_stack(2, 80, torch.float64, seed=10)
 
     def dense_h(V, Tinv):
         T = torch.linalg.solve_triangular(
