@@ -446,7 +446,7 @@ def _worker_world8_algorithms(rank, world, tmpfile, name):
             comm.broadcast(p.grad.data, src=0)
             torch.testing.assert_close(mine, p.grad,
                                        rtol=1e-4, atol=1e-5)
-    if name == "eigen":
+    if name == "eigen" and world > 4:
         # world 8 > 3 modules -> factor-wise: rank_g = rank_a + 1
         ras = sorted(ra for ra, _ in pre.module_ranks.values())
         rgs = [rg for _, rg in pre.module_ranks.values()]
@@ -464,6 +464,19 @@ def test_world8_all_algorithms(name):
     os.unlink(tmpfile)
     mp.spawn(_worker_world8_algorithms, args=(8, tmpfile, name),
              nprocs=8, join=True)
+
+
+@pytest.mark.parametrize("name", ["eigen", "eigen_dp", "inverse",
+                                  "inverse_dp"])
+def test_world3_odd_world_all_algorithms(name):
+    """Odd world size, and world == #modules (the layer-scheduling and
+    rotating-group modular arithmetic edge the power-of-two SCALE
+    widths never hit)."""
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(_worker_world8_algorithms, args=(3, tmpfile, name),
+             nprocs=3, join=True)
 
 
 def _worker_world8_inverse_modes(rank, world, tmpfile):
