@@ -190,3 +190,57 @@ def test_grouped_state_checkpoints(single_comm):
     load_kfac_state_dict(pre2, sd)
     torch.testing.assert_close(pre2.m_A[gm2.gc], pre.m_A[gm.gc])
     assert pre2.m_A[gm2.gc].dim() == 3
+
+
+# ------------------------------------------------------------- distributed
+def _worker_grouped_world2(rank, world, tmpfile, name):
+    """Grouped-conv model under a real multi-rank step: owner-computed
+    per-group factor blocks must broadcast/precondition identically on
+    every rank (grouped bucket shapes are (g, d, d) -- the 3-D case of
+    the owner-bucket comm path)."""
+    import torch.distributed as dist
+    import kfac_pytorch_amd as kfac
+    import kfac_pytorch_amd.parallel.comm as comm_mod
+    dist.init_process_group("gloo", init_method=f"file://{tmpfile}",
+                            world_size=world, rank=rank)
+    comm_mod.reset()
+    comm = comm_mod.init("Torch")
+    torch.manual_seed(17)
+
+    model = nn.Sequential(
+        nn.Conv2d(4, 8, 3, padding=1), nn.ReLU(),
+        nn.Conv2d(8, 8, 3, padding=1, groups=8, bias=True), nn.ReLU(),
+        nn.Conv2d(8, 12, 3, padding=1, groups=2), nn.ReLU(),
+        nn.Flatten(), nn.Linear(12 * 6 * 6, 5),
+    )
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+    pre = kfac.get_kfac_module(name)(model, damping=0.01)
+    g = torch.Generator().manual_seed(100 + rank)
+    x = torch.randn(6, 4, 6, 6, generator=g)
+    y = torch.randint(0, 5, (6,), generator=g)
+    for _ in range(2):
+        model.zero_grad()
+        F.cross_entropy(model(x), y).backward()
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        pre.step()
+        for p in model.parameters():
+            mine = p.grad.clone()
+            comm.broadcast(p.grad.data, src=0)
+            torch.testing.assert_close(mine, p.grad, rtol=1e-4,
+                                       atol=1e-5)
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("name", ["eigen", "eigen_dp", "inverse",
+                                  "inverse_dp"])
+def test_grouped_conv_world2_consistency(name):
+    import os
+    import tempfile
+    import torch.multiprocessing as mp
+    with tempfile.NamedTemporaryFile(delete=False) as f:
+        tmpfile = f.name
+    os.unlink(tmpfile)
+    mp.spawn(_worker_grouped_world2, args=(2, tmpfile, name), nprocs=2,
+             join=True)
