@@ -395,29 +395,37 @@ def test_world4_eigen_consistency():
 
 
 # --------------------------------------------------------------------------
-def _worker_exclude_comm_parts(rank, world, tmpfile):
-    """exclude_parts CommunicateFactor / CommunicateInverse at world 2:
-    the phases must be skipped without deadlock (reference ablation
-    harness: kfac_preconditioner_base.py:96-99,200-225)."""
+def _worker_exclude_comm_parts(rank, world, tmpfile, name, parts):
+    """exclude_parts ablations at world 2: every skipped phase must
+    skip identically on all ranks -- no deadlock, finite grads
+    (reference ablation harness:
+    kfac_preconditioner_base.py:96-99,200-225)."""
     import kfac_pytorch_amd as kfac
     comm = _init_worker(rank, world, tmpfile)
     torch.manual_seed(13)
     model = MLP()
     for p in model.parameters():
         comm.broadcast(p.data, src=0)
-    pre = kfac.KFAC_EIGEN(
-        model, damping=0.01,
-        exclude_parts='CommunicateFactor,CommunicateInverse')
+    pre = kfac.get_kfac_module(name)(model, damping=0.01,
+                                     exclude_parts=parts)
     x, y = _global_batch(seed=8)
     for _ in range(2):
         _train_grads(model, x, y)
-        pre.step()  # must not hang despite skipped collectives
+        pre.step()  # must not hang despite skipped phases
     assert all(torch.isfinite(p.grad).all() for p in model.parameters())
     dist.destroy_process_group()
 
 
-def test_exclude_comm_parts_no_deadlock():
-    _run_spawn(_worker_exclude_comm_parts)
+@pytest.mark.parametrize("name,parts", [
+    ("eigen", "CommunicateFactor,CommunicateInverse"),
+    ("eigen", "ComputeInverse"),
+    ("inverse", "CommunicateInverse"),
+    ("inverse_dp", "ComputeInverse"),
+    ("eigen_dp", "ComputeInverse"),
+    ("eigen_dp", "ComputeFactor"),
+])
+def test_exclude_parts_no_deadlock(name, parts):
+    _run_spawn(_worker_exclude_comm_parts, args=(name, parts))
 
 
 # --------------------------------------------------------------------------
