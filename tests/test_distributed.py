@@ -599,3 +599,56 @@ def _worker_lpt_balance(rank, world, tmpfile):
 
 def test_lpt_schedule_balances_and_agrees():
     _run_spawn(_worker_lpt_balance)
+
+
+# --------------------------------------------------------------------------
+def _worker_vocab_and_scheduler(rank, world, tmpfile):
+    """Vocab exclusion + KFACParamScheduler under multi-rank: both
+    ranks must register the SAME module set (a mismatch deadlocks the
+    owner broadcasts) and stay grad-consistent across a frequency
+    boundary driven by the scheduler."""
+    import kfac_pytorch_amd as kfac
+    comm = _init_worker(rank, world, tmpfile)
+    torch.manual_seed(29)
+
+    class LM(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc1 = nn.Linear(6, 8)
+            self.head = nn.Linear(8, 211)   # "vocab"-sized projection
+
+        def forward(self, x):
+            return self.head(F.relu(self.fc1(x)))
+
+    model = LM()
+    for p in model.parameters():
+        comm.broadcast(p.data, src=0)
+    pre = kfac.KFAC_EIGEN_DP(model, damping=0.01,
+                             exclude_vocabulary_size=211,
+                             fac_update_freq=1, kfac_update_freq=1)
+    # the head must be excluded identically everywhere
+    assert len(pre.modules) == 1
+    sched = kfac.KFACParamScheduler(pre, damping_alpha=0.5,
+                                    damping_schedule=[1],
+                                    update_freq_alpha=2,
+                                    update_freq_schedule=[1])
+    g = torch.Generator().manual_seed(300 + rank)
+    x = torch.randn(8, 6, generator=g)
+    y = torch.randint(0, 211, (8,), generator=g)
+    for epoch in range(3):
+        sched.step(epoch)
+        _train_grads(model, x, y)
+        for p in model.parameters():
+            comm.allreduce(p.grad.data, op=comm.Average)
+        pre.step()
+        for p in model.parameters():
+            mine = p.grad.clone()
+            comm.broadcast(p.grad.data, src=0)
+            torch.testing.assert_close(mine, p.grad, rtol=1e-4,
+                                       atol=1e-5)
+    assert pre.fac_update_freq >= 1 and pre.kfac_update_freq >= 1
+    dist.destroy_process_group()
+
+
+def test_vocab_exclusion_and_scheduler_world2():
+    _run_spawn(_worker_vocab_and_scheduler)
