@@ -270,7 +270,65 @@ def bulge_chase_batched(B: torch.Tensor, b: int
     return A, Q2
 
 
-def sbr_eigh_batched(A: torch.Tensor, b: int = 64
+def bulge_chase_blocked_batched(B: torch.Tensor, b: int
+                                ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Stage 2, BLOCKED form: band(b) -> tridiagonal with length-b
+    Householder reflectors -- one reflector eliminates column j's band
+    below the subdiagonal, then each hop QRs the filled b x b bulge
+    block and pushes it b rows down (docs/SBR_STAGE2_NOTES.md).  This
+    is the exact per-hop math the wavefront HIP kernel runs (the
+    per-element :func:`bulge_chase_batched` is its Givens oracle: both
+    fix e1, so by the implicit-Q theorem their tridiagonals agree up
+    to signs).  Returns ``(T, Q2)`` with ``B = Q2 T Q2^T``.
+
+    Sequential depth per sweep is ~n/b reflector hops instead of
+    ~n per-element rotations -- the property that makes one grid
+    barrier per macro-step feasible on the GPU.
+    """
+    A = B.clone()
+    N, n, _ = A.shape
+    Q2 = torch.eye(n, dtype=A.dtype, device=A.device) \
+        .expand(N, n, n).contiguous()
+
+    def apply_block(rs, re, cs, ce):
+        """QR the (rs:re, cs:ce) block in-place (rows rs:re mix) and
+        apply the block reflector two-sided + into Q2."""
+        P = A[:, rs:re, cs:ce].contiguous()
+        M, k = P.shape[1], min(P.shape[1], P.shape[2])
+        if M < 2:
+            return
+        V, Tinv, R = _panel_wy_geqrf(P)
+        # rows: A[S,:] <- H^T A[S,:]
+        W = torch.bmm(V.mT, A[:, rs:re, :])
+        TtW = torch.linalg.solve_triangular(Tinv.mT, W, upper=False)
+        A[:, rs:re, :] -= torch.bmm(V, TtW)
+        # cols: A[:,S] <- A[:,S] H   (and the same for Q2)
+        for Mt in (A, Q2):
+            Y = torch.bmm(Mt[:, :, rs:re], V)
+            YT = torch.linalg.solve_triangular(Tinv, Y, upper=True,
+                                               left=False)
+            Mt[:, :, rs:re] -= torch.bmm(YT, V.mT)
+        # exact zeros on the annihilated block (and its mirror): the
+        # top k rows hold H^T P = R
+        A[:, rs:re, cs:ce] = 0.0
+        A[:, rs:rs + k, cs:ce] = R
+        A[:, cs:ce, rs:re] = 0.0
+        A[:, cs:ce, rs:rs + k] = R.mT
+
+    for j in range(n - 2):
+        s, e = j + 1, min(j + 1 + b, n)
+        if e - s >= 2:
+            apply_block(s, e, j, j + 1)
+        while e < n:
+            ns, ne = e, min(e + b, n)
+            apply_block(ns, ne, s, e)
+            s, e = ns, ne
+    A = 0.5 * (A + A.mT)
+    return A, Q2
+
+
+def sbr_eigh_batched(A: torch.Tensor, b: int = 64,
+                     chase: str = "blocked"
                      ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Complete two-stage SBR symmetric eigensolve, batched:
     full -> band(b) -> tridiagonal -> eigh, with both back-transforms.
@@ -279,12 +337,15 @@ def sbr_eigh_batched(A: torch.Tensor, b: int = 64
 
     End-to-end correctness pipeline for the round-3 kernel: stage 1
     runs at batched-GEMM rate (:func:`band_reduce_batched`), stage 2
-    is the per-element reference chase (the HIP kernel replaces
-    exactly that inner loop), the tridiagonal solve stands in for the
-    existing stedc binding.  Tested against ``torch.linalg.eigh``.
+    is the reflector-blocked chase (``chase="blocked"``, the form the
+    HIP kernel implements hop for hop) or the per-element Givens
+    oracle (``chase="givens"``); the tridiagonal solve stands in for
+    the existing stedc binding.  Tested against ``torch.linalg.eigh``.
     """
     Bb, panels = band_reduce_batched(A, b)
-    T, Q2 = bulge_chase_batched(Bb, b)
+    chase_fn = {"blocked": bulge_chase_blocked_batched,
+                "givens": bulge_chase_batched}[chase]
+    T, Q2 = chase_fn(Bb, b)
     d, Z = torch.linalg.eigh(T)
     vecs = apply_q_batched(panels, torch.bmm(Q2, Z))
     return d, vecs

@@ -204,3 +204,38 @@ def test_sbr_eigh_rank_deficient_fp32():
     assert float(resid) < 1e-4
     ev = torch.linalg.eigvalsh(A)
     assert float((d - ev).abs().max() / ev.abs().max()) < 1e-4
+
+
+@pytest.mark.parametrize("n,b", [(24, 4), (48, 8), (65, 8), (40, 16)])
+def test_blocked_chase_matches_givens_oracle(n, b):
+    """The reflector-blocked chase (the HIP kernel's exact per-hop
+    math) against the per-element Givens oracle: both fix e1, so the
+    implicit-Q theorem forces the same tridiagonal up to signs."""
+    from kfac_pytorch_amd.ops.sbr import (band_reduce_batched,
+                                          bulge_chase_batched,
+                                          bulge_chase_blocked_batched)
+    A = _spd_stack(3, n, torch.float64, seed=n + 3)
+    B, _ = band_reduce_batched(A, b)
+    T1, Q1 = bulge_chase_blocked_batched(B, b)
+    assert float(T1.triu(2).abs().max()) == 0.0
+    eye = torch.eye(n, dtype=torch.float64)
+    assert float((Q1 @ Q1.mT - eye).abs().max()) < 1e-12
+    resid = (Q1 @ T1 @ Q1.mT - B).norm() / B.norm()
+    assert float(resid) < 1e-13
+    T2, _ = bulge_chase_batched(B, b)
+    dd = (T1.diagonal(dim1=-2, dim2=-1)
+          - T2.diagonal(dim1=-2, dim2=-1)).abs().max()
+    od = (T1.diagonal(offset=1, dim1=-2, dim2=-1).abs()
+          - T2.diagonal(offset=1, dim1=-2, dim2=-1).abs()).abs().max()
+    assert float(dd) < 1e-10 and float(od) < 1e-10
+
+
+@pytest.mark.parametrize("chase", ["blocked", "givens"])
+def test_sbr_eigh_both_chases(chase):
+    from kfac_pytorch_amd.ops.sbr import sbr_eigh_batched
+    A = _spd_stack(2, 48, torch.float64, seed=8)
+    d, V = sbr_eigh_batched(A, 8, chase=chase)
+    resid = (A @ V - V * d.unsqueeze(-2)).abs().max()
+    assert float(resid) < 1e-12
+    ev = torch.linalg.eigvalsh(A)
+    assert float((d - ev).abs().max() / ev.abs().max()) < 1e-12
