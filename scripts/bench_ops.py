@@ -106,10 +106,29 @@ def bench_im2col():
               f"unfold {t_unf * 1e3:7.2f} ms")
 
 
+def bench_sbr():
+    """SBR stage-1 band reduction vs whole-solve library time at the
+    big buckets (the round-3 eigensolver path; stage-2 chase numbers
+    need the HIP kernel -- the torch chase is a correctness oracle,
+    not a perf path).  NOT in --op all: opt in explicitly."""
+    from kfac_pytorch_amd.ops.sbr import band_reduce_batched
+    print("== SBR stage 1 (full->band 64) vs library whole solve ==")
+    for n, bsz in [(1152, 4), (2304, 4), (4608, 3)]:
+        g = torch.Generator().manual_seed(n)
+        x = torch.randn(bsz, n, 2 * n, generator=g)
+        A = (x @ x.mT / (2 * n)).cuda()
+        t1 = timeit(lambda: band_reduce_batched(A, 64), warmup=2,
+                    iters=3)
+        t2 = timeit(lambda: torch.linalg.eigh(A), warmup=1, iters=2)
+        print(f"n={n:5d}x{bsz}  stage1 {t1 * 1e3:8.1f} ms   "
+              f"eigh(whole) {t2 * 1e3:8.1f} ms")
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--op", default="all",
-                   choices=["all", "eig", "syrk", "precond", "im2col"])
+                   choices=["all", "eig", "syrk", "precond", "im2col",
+                            "sbr"])
     args = p.parse_args()
     assert torch.cuda.is_available(), "bench_ops needs a GPU"
     if args.op in ("all", "eig"):
@@ -120,6 +139,8 @@ def main():
         bench_precond()
     if args.op in ("all", "im2col"):
         bench_im2col()
+    if args.op == "sbr":
+        bench_sbr()
 
 
 if __name__ == "__main__":
