@@ -1,17 +1,23 @@
-"""Two-stage SBR tridiagonalization, stage 1: batched full -> band
-reduction at GEMM rate (docs/SBR_STAGE2_NOTES.md; numpy oracle
-scripts/sbr_ref.py, validated to machine precision).
+"""Two-stage SBR tridiagonalization, batched end to end
+(docs/SBR_STAGE2_NOTES.md; numpy oracle scripts/sbr_ref.py, validated
+to machine precision).
 
-This is the round-3 eigensolver path's first half, shipped as a
-working, tested torch implementation: every panel is annihilated with
-ONE batched ``geqrf`` (no per-column sequencing -- the measured cost
-of every one-stage tridiagonalization, docs/SYTRD_DESIGN.md), and the
+Stage 1 (full -> band b): each panel is annihilated with ONE batched
+panel factorization (no per-column sequencing -- the measured cost of
+every one-stage tridiagonalization, docs/SYTRD_DESIGN.md), and the
 two-sided update runs as four batched GEMMs in compact-WY form using
 the same ``T^{-1} = diag(1/tau) + strict_upper(V^T V)`` identity as
-``ops/linalg.py::_wy_backtransform``.  The remaining half (band ->
-tridiagonal bulge chase, reference analog of rocSOLVER's internal
-sytrd path) is the round-3 HIP kernel; until it exists this module is
-not wired into ``mat_eig_multi`` and carries no env flag.
+``ops/linalg.py::_wy_backtransform``.  Measured on MI355X
+(profiles/sbr_stage1_deferred.log).
+
+Stage 2 (band -> tridiagonal): TWO validated chase implementations --
+the per-element Givens oracle and the reflector-blocked chase the
+wavefront HIP kernel runs hop for hop (both fix e1, so implicit-Q
+forces the same tridiagonal up to signs).  The torch chases are
+correctness references: sequential at Python speed, so this module is
+not wired into ``mat_eig_multi`` -- the next-round kernel replaces
+exactly the chase inner loop of :func:`sbr_eigh_batched`, which
+already passes eigh-parity tests batched.
 
 Reference analog: the ``tcmm_symeig`` replacement contract
 (/root/reference/packages/tcmm/src/tcmm_kernel.cu:56-116) -- this is
@@ -25,7 +31,9 @@ from typing import List, Tuple
 
 import torch
 
-__all__ = ["band_reduce_batched", "apply_q_batched"]
+__all__ = ["band_reduce_batched", "apply_q_batched",
+           "bulge_chase_batched", "bulge_chase_blocked_batched",
+           "sbr_eigh_batched"]
 
 Panel = Tuple[int, torch.Tensor, torch.Tensor]  # (r0, V, Tinv)
 
