@@ -7,6 +7,9 @@
 set -e
 N=${1:-8}
 cd "$(dirname "$0")/.."
+# dmabuf IPC is the only mode this driver stack supports; RCCL fails
+# with hipIpcGetMemHandle errors without it
+export HSA_ENABLE_IPC_MODE_LEGACY=0
 RUN="python -m torch.distributed.run --nnodes=1 --nproc-per-node $N \
      --master-addr 127.0.0.1 --master-port 29755 bench.py --gpus $N \
      --steps 5 --warmup 2"
