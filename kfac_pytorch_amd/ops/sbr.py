@@ -297,42 +297,57 @@ def bulge_chase_blocked_batched(B: torch.Tensor, b: int
     N, n, _ = A.shape
     Q2 = torch.eye(n, dtype=A.dtype, device=A.device) \
         .expand(N, n, n).contiguous()
-
-    def apply_block(rs, re, cs, ce):
-        """QR the (rs:re, cs:ce) block in-place (rows rs:re mix) and
-        apply the block reflector two-sided + into Q2."""
-        P = A[:, rs:re, cs:ce].contiguous()
-        M, k = P.shape[1], min(P.shape[1], P.shape[2])
-        if M < 2:
-            return
-        V, Tinv, R = _panel_wy_geqrf(P)
-        # rows: A[S,:] <- H^T A[S,:]
-        W = torch.bmm(V.mT, A[:, rs:re, :])
-        TtW = torch.linalg.solve_triangular(Tinv.mT, W, upper=False)
-        A[:, rs:re, :] -= torch.bmm(V, TtW)
-        # cols: A[:,S] <- A[:,S] H   (and the same for Q2)
-        for Mt in (A, Q2):
-            Y = torch.bmm(Mt[:, :, rs:re], V)
-            YT = torch.linalg.solve_triangular(Tinv, Y, upper=True,
-                                               left=False)
-            Mt[:, :, rs:re] -= torch.bmm(YT, V.mT)
-        # exact zeros on the annihilated block (and its mirror): the
-        # top k rows hold H^T P = R
-        A[:, rs:re, cs:ce] = 0.0
-        A[:, rs:rs + k, cs:ce] = R
-        A[:, cs:ce, rs:re] = 0.0
-        A[:, cs:ce, rs:rs + k] = R.mT
-
     for j in range(n - 2):
-        s, e = j + 1, min(j + 1 + b, n)
-        if e - s >= 2:
-            apply_block(s, e, j, j + 1)
-        while e < n:
-            ns, ne = e, min(e + b, n)
-            apply_block(ns, ne, s, e)
-            s, e = ns, ne
+        for _ in _sweep_hops(A, Q2, j, b):
+            pass
     A = 0.5 * (A + A.mT)
     return A, Q2
+
+
+def _apply_chase_block(A, Q2, rs, re, cs, ce):
+    """QR the (rs:re, cs:ce) block in-place (rows rs:re mix) and apply
+    the block reflector two-sided + into Q2."""
+    P = A[:, rs:re, cs:ce].contiguous()
+    M, k = P.shape[1], min(P.shape[1], P.shape[2])
+    if M < 2:
+        return
+    V, Tinv, R = _panel_wy_geqrf(P)
+    # rows: A[S,:] <- H^T A[S,:]
+    W = torch.bmm(V.mT, A[:, rs:re, :])
+    TtW = torch.linalg.solve_triangular(Tinv.mT, W, upper=False)
+    A[:, rs:re, :] -= torch.bmm(V, TtW)
+    # cols: A[:,S] <- A[:,S] H   (and the same for Q2)
+    for Mt in (A, Q2):
+        Y = torch.bmm(Mt[:, :, rs:re], V)
+        YT = torch.linalg.solve_triangular(Tinv, Y, upper=True,
+                                           left=False)
+        Mt[:, :, rs:re] -= torch.bmm(YT, V.mT)
+    # exact zeros on the annihilated block (and its mirror): the top k
+    # rows hold H^T P = R
+    A[:, rs:re, cs:ce] = 0.0
+    A[:, rs:rs + k, cs:ce] = R
+    A[:, cs:ce, rs:re] = 0.0
+    A[:, cs:ce, rs:rs + k] = R.mT
+
+
+def _sweep_hops(A, Q2, j, b):
+    """Generator over sweep ``j``'s reflector hops (mutates A/Q2 in
+    place, yields after each hop).  Hop t's row window is
+    ``(j + t*b, min(j + (t+1)*b, n)]`` -- windows of sweeps whose hop
+    indices differ by >= 2 are disjoint, which is the wavefront
+    pipelining precondition the HIP kernel relies on
+    (tests/test_sbr_stage1.py::test_wavefront_interleaving_commutes
+    validates it by driving these generators interleaved)."""
+    n = A.shape[-1]
+    s, e = j + 1, min(j + 1 + b, n)
+    if e - s >= 2:
+        _apply_chase_block(A, Q2, s, e, j, j + 1)
+    yield
+    while e < n:
+        ns, ne = e, min(e + b, n)
+        _apply_chase_block(A, Q2, ns, ne, s, e)
+        s, e = ns, ne
+        yield
 
 
 def sbr_eigh_batched(A: torch.Tensor, b: int = 64,

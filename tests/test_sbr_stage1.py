@@ -239,3 +239,38 @@ def test_sbr_eigh_both_chases(chase):
     assert float(resid) < 1e-12
     ev = torch.linalg.eigvalsh(A)
     assert float((d - ev).abs().max() / ev.abs().max()) < 1e-12
+
+
+def test_wavefront_interleaving_commutes():
+    """The HIP kernel's pipelining precondition: sweeps whose hop
+    indices stay >= 2 apart act on disjoint row windows, so the
+    interleaved (wavefront) schedule must produce the same
+    tridiagonal as the sequential chase (exactly, in exact
+    arithmetic; to roundoff here)."""
+    import kfac_pytorch_amd.ops.sbr as sbr
+    n, b = 48, 8
+    A0 = _spd_stack(2, n, torch.float64, seed=31)
+    B, _ = sbr.band_reduce_batched(A0, b)
+
+    T_seq, Q_seq = sbr.bulge_chase_blocked_batched(B, b)
+
+    A = B.clone()
+    Q2 = torch.eye(n, dtype=A.dtype).expand(2, n, n).contiguous()
+    gens, hops, nxt = {}, {}, 0
+    while gens or nxt < n - 2:
+        if nxt < n - 2 and (nxt == 0 or hops.get(nxt - 1, 0) >= 2
+                            or (nxt - 1) not in gens):
+            gens[nxt] = sbr._sweep_hops(A, Q2, nxt, b)
+            hops[nxt] = 0
+            nxt += 1
+        for j in sorted(gens):
+            try:
+                next(gens[j])
+                hops[j] += 1
+            except StopIteration:
+                del gens[j]
+    T_wave = 0.5 * (A + A.mT)
+
+    assert float(T_wave.triu(2).abs().max()) < 1e-12
+    torch.testing.assert_close(T_wave, T_seq, rtol=1e-9, atol=1e-9)
+    torch.testing.assert_close(Q2, Q_seq, rtol=1e-9, atol=1e-9)
