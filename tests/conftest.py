@@ -44,3 +44,21 @@ def single_process_comm():
 def seeded():
     torch.manual_seed(1234)
     yield
+
+
+def pytest_runtest_logreport(report):
+    """Append every failure to tests/.failures.log (name, phase, the
+    first error line) -- rare environment-correlated flakes otherwise
+    vanish with the terminal scrollback."""
+    if report.failed:
+        import datetime
+        import os
+        path = os.path.join(os.path.dirname(__file__), ".failures.log")
+        first = ""
+        try:
+            first = str(report.longrepr).splitlines()[-1][:300]
+        except Exception:
+            pass
+        with open(path, "a") as f:
+            f.write(f"{datetime.datetime.now().isoformat()} "
+                    f"{report.nodeid} [{report.when}] {first}\n")
