@@ -29,11 +29,22 @@ def _init_worker(rank, world_size, tmpfile):
     return comm_mod.get_comm()
 
 
-def _run_spawn(fn, args=()):
+def _tmpfile():
     with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(fn, args=(WORLD, tmpfile) + args, nprocs=WORLD, join=True)
+        name = f.name
+    os.unlink(name)
+    return name
+
+
+def _spawn(fn, world, extra=()):
+    """mp.spawn with one environment-flake retry on a FRESH file-store
+    rendezvous (tests/conftest.py::spawn_retry)."""
+    from tests.conftest import spawn_retry
+    spawn_retry(fn, lambda: (world, _tmpfile()) + tuple(extra), world)
+
+
+def _run_spawn(fn, args=()):
+    _spawn(fn, WORLD, args)
 
 
 class MLP(nn.Module):
@@ -388,10 +399,7 @@ def _worker_world4_eigen(rank, world, tmpfile):
 
 
 def test_world4_eigen_consistency():
-    with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(_worker_world4_eigen, args=(4, tmpfile), nprocs=4, join=True)
+    _spawn(_worker_world4_eigen, 4)
 
 
 # --------------------------------------------------------------------------
@@ -467,11 +475,7 @@ def _worker_world8_algorithms(rank, world, tmpfile, name):
 @pytest.mark.parametrize("name", ["eigen", "eigen_dp", "inverse",
                                   "inverse_dp"])
 def test_world8_all_algorithms(name):
-    with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(_worker_world8_algorithms, args=(8, tmpfile, name),
-             nprocs=8, join=True)
+    _spawn(_worker_world8_algorithms, 8, (name,))
 
 
 @pytest.mark.parametrize("name", ["eigen", "eigen_dp", "inverse",
@@ -480,11 +484,7 @@ def test_world3_odd_world_all_algorithms(name):
     """Odd world size, and world == #modules (the layer-scheduling and
     rotating-group modular arithmetic edge the power-of-two SCALE
     widths never hit)."""
-    with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(_worker_world8_algorithms, args=(3, tmpfile, name),
-             nprocs=3, join=True)
+    _spawn(_worker_world8_algorithms, 3, (name,))
 
 
 def _worker_world8_inverse_modes(rank, world, tmpfile):
@@ -512,11 +512,7 @@ def _worker_world8_inverse_modes(rank, world, tmpfile):
 
 
 def test_world8_inverse_both_comm_modes():
-    with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(_worker_world8_inverse_modes, args=(8, tmpfile),
-             nprocs=8, join=True)
+    _spawn(_worker_world8_inverse_modes, 8)
 
 
 def _worker_rotating_fewer_than_world(rank, world, tmpfile):
@@ -542,11 +538,7 @@ def _worker_rotating_fewer_than_world(rank, world, tmpfile):
 
 
 def test_rotating_groups_fewer_than_world():
-    with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(_worker_rotating_fewer_than_world, args=(4, tmpfile),
-             nprocs=4, join=True)
+    _spawn(_worker_rotating_fewer_than_world, 4)
 
 
 def _worker_lpt_balance(rank, world, tmpfile):

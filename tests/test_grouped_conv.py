@@ -238,9 +238,12 @@ def _worker_grouped_world2(rank, world, tmpfile, name):
 def test_grouped_conv_world2_consistency(name):
     import os
     import tempfile
-    import torch.multiprocessing as mp
-    with tempfile.NamedTemporaryFile(delete=False) as f:
-        tmpfile = f.name
-    os.unlink(tmpfile)
-    mp.spawn(_worker_grouped_world2, args=(2, tmpfile, name), nprocs=2,
-             join=True)
+    from tests.conftest import spawn_retry
+
+    def tmpfile():
+        with tempfile.NamedTemporaryFile(delete=False) as f:
+            name_ = f.name
+        os.unlink(name_)
+        return name_
+
+    spawn_retry(_worker_grouped_world2, lambda: (2, tmpfile(), name), 2)
