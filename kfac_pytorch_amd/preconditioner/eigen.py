@@ -105,7 +105,7 @@ class EigenComputeMixin:
         tol = float(os.environ.get("KFAC_WARM_TOL", "5e-6"))
         max_iters = int(os.environ.get("KFAC_WARM_MAX_ITERS", "12"))
         if not hasattr(self, "_warm_state"):
-            self._warm_state = {}  # (id(mod), kind) -> [V, age]
+            self._warm_state = {}  # (id(mod), kind, gi) -> [V, age]
         results = [None] * len(mats)
         groups = {}
         for i, (mod, kind, gi) in enumerate(work):
